@@ -1,0 +1,21 @@
+import os
+import sys
+
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires a real MI355X (run with -m gpu)")
+
+
+def pytest_collection_modifyitems(config, items):
+    # `-m gpu` / `-m "not gpu"` handles selection; nothing extra needed here.
+    pass
+
+
+@pytest.fixture(scope="session")
+def golden_dir():
+    return os.path.join(os.path.dirname(os.path.abspath(__file__)), "golden")
