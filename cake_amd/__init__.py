@@ -1,0 +1,269 @@
+"""cake_amd — MI355X-native engine for cake's layer-sharded LLM hot path.
+
+Host-side mirror of the reference's operator surface (ComputeBackend,
+backends/mod.rs:40-685; Forwarder, cake/mod.rs:511-556; Topology YAML,
+sharding/topology.rs) over the C-ABI of include/cake_hip.h.
+
+The product path is the HIP engine in libcake_hip.so (hand-written gfx950
+kernels + RCCL over xGMI).  There is NO CPU fallback: if the extension is
+missing, importing this package raises, and every compute call requires a
+real GPU.  The CPU oracle lives in oracle/ and is test infrastructure only.
+"""
+import ctypes
+import json
+import os
+
+import numpy as np
+
+_PKG_DIR = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_PKG_DIR, "libcake_hip.so")
+
+COMM_ID_BYTES = 128
+HAS_EMBED = 1
+HAS_HEAD = 2
+USE_GRAPH = 4
+STATS = 8
+
+
+class CakeHipError(RuntimeError):
+    pass
+
+
+def _load_lib():
+    if not os.path.exists(_LIB_PATH):
+        raise ImportError(
+            f"cake_amd: native engine not built ({_LIB_PATH} missing). "
+            "Run __graft_entry__.build() — there is no CPU fallback.")
+    lib = ctypes.CDLL(_LIB_PATH)
+    c = ctypes.c_int
+    p = ctypes.c_void_p
+    cp = ctypes.c_char_p
+    fp = ctypes.POINTER(ctypes.c_float)
+    u32p = ctypes.POINTER(ctypes.c_uint32)
+    lib.cake_hip_last_error.restype = cp
+    lib.cake_hip_build_info.restype = cp
+    lib.cake_hip_engine_create.argtypes = [cp, c, c, c, c, c, c,
+                                           ctypes.POINTER(p)]
+    lib.cake_hip_engine_free.argtypes = [p]
+    lib.cake_hip_topology_node_range.argtypes = [cp, cp, ctypes.POINTER(c),
+                                                 ctypes.POINTER(c)]
+    lib.cake_hip_load_safetensors.argtypes = [p, cp]
+    lib.cake_hip_init_random.argtypes = [p, ctypes.c_uint64, ctypes.c_float]
+    lib.cake_hip_prefill.argtypes = [p, u32p, c, u32p, fp]
+    lib.cake_hip_decode.argtypes = [p, c, u32p]
+    lib.cake_hip_reset.argtypes = [p]
+    lib.cake_hip_forward_hidden.argtypes = [p, fp, c, c, fp]
+    lib.cake_hip_comm_id.argtypes = [ctypes.c_char_p]
+    lib.cake_hip_comm_init.argtypes = [p, c, c, ctypes.c_char_p]
+    lib.cake_hip_op_rms_norm.argtypes = [c, c, ctypes.c_float, fp, fp, fp, c]
+    lib.cake_hip_op_linear.argtypes = [c, c, c, fp, fp, fp, c]
+    lib.cake_hip_op_silu_mul.argtypes = [ctypes.c_long, fp, fp, fp, c]
+    lib.cake_hip_op_rope.argtypes = [c, c, c, c, fp, fp, fp, fp, c]
+    lib.cake_hip_kernel_stats.argtypes = [p, ctypes.c_char_p, c]
+    lib.cake_hip_stats_reset.argtypes = [p]
+    lib.cake_hip_set_stats.argtypes = [p, c]
+    lib.cake_hip_sync.argtypes = [p]
+    return lib
+
+
+_lib = _load_lib()
+
+
+def _check(code):
+    if code != 0:
+        raise CakeHipError(_lib.cake_hip_last_error().decode())
+
+
+def build_info():
+    return _lib.cake_hip_build_info().decode()
+
+
+def _f32(a):
+    a = np.ascontiguousarray(a, dtype=np.float32)
+    return a, a.ctypes.data_as(ctypes.POINTER(ctypes.c_float))
+
+
+def topology_node_range(yaml_text: str, node: str):
+    """Expand a cake topology YAML (topology.rs:134-169, incl. range
+    expressions) to node's contiguous layer range [lo, hi)."""
+    lo = ctypes.c_int()
+    hi = ctypes.c_int()
+    _check(_lib.cake_hip_topology_node_range(
+        yaml_text.encode(), node.encode(), ctypes.byref(lo),
+        ctypes.byref(hi)))
+    return lo.value, hi.value
+
+
+# ---------------------------------------------------------------------------
+# op-level surface (kernel parity tests)
+# ---------------------------------------------------------------------------
+def op_rms_norm(x, w, eps=1e-5, device=0):
+    x = np.ascontiguousarray(x, dtype=np.float32)
+    rows = int(np.prod(x.shape[:-1])) if x.ndim > 1 else 1
+    cols = x.shape[-1]
+    out = np.empty_like(x)
+    _, xp = _f32(x)
+    _, wp = _f32(w)
+    _check(_lib.cake_hip_op_rms_norm(
+        rows, cols, ctypes.c_float(eps), xp, wp,
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), device))
+    return out
+
+
+def op_linear(x, w, device=0):
+    """x (M,K) @ w(N,K)^T -> (M,N)"""
+    x = np.ascontiguousarray(x, dtype=np.float32)
+    w = np.ascontiguousarray(w, dtype=np.float32)
+    M, K = x.shape
+    N = w.shape[0]
+    out = np.empty((M, N), dtype=np.float32)
+    _, xp = _f32(x)
+    _, wp = _f32(w)
+    _check(_lib.cake_hip_op_linear(
+        M, N, K, xp, wp, out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+        device))
+    return out
+
+
+def op_silu_mul(gate, up, device=0):
+    gate = np.ascontiguousarray(gate, dtype=np.float32)
+    out = np.empty_like(gate)
+    _, gp = _f32(gate)
+    _, up_ = _f32(up)
+    _check(_lib.cake_hip_op_silu_mul(
+        gate.size, gp, up_,
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), device))
+    return out
+
+
+def op_rope(x, cos, sin, device=0):
+    """x (B,H,S,D), cos/sin (S, D/2)"""
+    x = np.ascontiguousarray(x, dtype=np.float32)
+    b, h, s, d = x.shape
+    out = np.empty_like(x)
+    _, xp = _f32(x)
+    _, cp = _f32(cos)
+    _, sp = _f32(sin)
+    _check(_lib.cake_hip_op_rope(
+        b, h, s, d, xp, cp, sp,
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), device))
+    return out
+
+
+def comm_id() -> bytes:
+    buf = ctypes.create_string_buffer(COMM_ID_BYTES)
+    _check(_lib.cake_hip_comm_id(buf))
+    return buf.raw
+
+
+# ---------------------------------------------------------------------------
+# Engine — host mirror of Master/TextModelBase over the C-ABI
+# ---------------------------------------------------------------------------
+class Engine:
+    def __init__(self, config, layer_lo=0, layer_hi=None, flags=None,
+                 max_seq=0, max_batch_tokens=0, device=0):
+        if isinstance(config, dict):
+            config = json.dumps(config)
+        cfg = json.loads(config)
+        if layer_hi is None:
+            layer_hi = cfg["num_hidden_layers"]
+        if flags is None:
+            flags = HAS_EMBED | HAS_HEAD | USE_GRAPH
+        self.cfg = cfg
+        self.flags = flags
+        self.vocab = cfg["vocab_size"]
+        self.hidden = cfg["hidden_size"]
+        h = ctypes.c_void_p()
+        _check(_lib.cake_hip_engine_create(
+            config.encode(), layer_lo, layer_hi, flags, max_seq,
+            max_batch_tokens, device, ctypes.byref(h)))
+        self._h = h
+
+    @classmethod
+    def from_topology(cls, config, topology_yaml, node, **kw):
+        lo, hi = topology_node_range(topology_yaml, node)
+        return cls(config, layer_lo=lo, layer_hi=hi, **kw)
+
+    def load_safetensors(self, path):
+        _check(_lib.cake_hip_load_safetensors(self._h, path.encode()))
+
+    def init_random(self, seed=299792458, scale=0.02):
+        _check(_lib.cake_hip_init_random(self._h, seed, ctypes.c_float(scale)))
+
+    def comm_init(self, rank, world, comm_id_bytes):
+        _check(_lib.cake_hip_comm_init(self._h, rank, world, comm_id_bytes))
+
+    def prefill(self, tokens, want_logits=False):
+        tokens = np.ascontiguousarray(tokens, dtype=np.uint32)
+        nxt = ctypes.c_uint32()
+        logits = np.empty(self.vocab, dtype=np.float32) if want_logits else None
+        _check(_lib.cake_hip_prefill(
+            self._h,
+            tokens.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+            len(tokens), ctypes.byref(nxt),
+            logits.ctypes.data_as(ctypes.POINTER(ctypes.c_float))
+            if want_logits else None))
+        return (nxt.value, logits) if want_logits else nxt.value
+
+    def prefill_participate(self, n_tokens):
+        """Non-rank-0 side of a pipelined prefill."""
+        _check(_lib.cake_hip_prefill(self._h, None, n_tokens, None, None))
+
+    def decode(self, steps):
+        out = np.empty(steps, dtype=np.uint32)
+        _check(_lib.cake_hip_decode(
+            self._h, steps,
+            out.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32))))
+        return out
+
+    def decode_participate(self, steps):
+        _check(_lib.cake_hip_decode(self._h, steps, None))
+
+    def forward_hidden(self, x, index_pos=0):
+        """Run this shard's blocks on (seq, hidden) f32 hidden states —
+        the Forwarder::forward_batch unit (cake/mod.rs:533-540)."""
+        x = np.ascontiguousarray(x, dtype=np.float32)
+        seq = x.shape[0]
+        out = np.empty_like(x)
+        _check(_lib.cake_hip_forward_hidden(
+            self._h, x.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), seq,
+            index_pos, out.ctypes.data_as(ctypes.POINTER(ctypes.c_float))))
+        return out
+
+    def reset(self):
+        _check(_lib.cake_hip_reset(self._h))
+
+    def sync(self):
+        _check(_lib.cake_hip_sync(self._h))
+
+    def set_stats(self, enabled):
+        _check(_lib.cake_hip_set_stats(self._h, 1 if enabled else 0))
+
+    def stats_reset(self):
+        _check(_lib.cake_hip_stats_reset(self._h))
+
+    def kernel_stats(self):
+        buf = ctypes.create_string_buffer(1 << 20)
+        _check(_lib.cake_hip_kernel_stats(self._h, buf, len(buf)))
+        return json.loads(buf.value.decode())
+
+    def generate_greedy(self, prompt_ids, max_new):
+        """Greedy generation mirroring Master::generate_text's loop
+        (master.rs:109-171): prefill produces token 0, decode the rest."""
+        self.reset()
+        first = self.prefill(prompt_ids)
+        if max_new == 1:
+            return [first]
+        rest = self.decode(max_new - 1)
+        return [first] + list(rest)
+
+    def close(self):
+        if self._h:
+            _lib.cake_hip_engine_free(self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass

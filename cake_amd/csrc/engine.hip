@@ -1,0 +1,1182 @@
+// cake_hip engine — MI355X-native host runtime for cake's layer-sharded LLM
+// hot path, behind the C-ABI of include/cake_hip.h.
+//
+// Replaces (from scratch, not a port):
+//   - Context::from_args + TextModelBase::load  (cake/mod.rs:114-507,
+//     text_model.rs:150-263): config.json parse, safetensors load with
+//     per-shard tensor selection, fused QKV / gate_up weights
+//     (attention.rs:109-114, mlp.rs:38-46), direct HBM upload as bf16
+//   - the generation loop (text_model.rs:266-368,397-495): greedy ArgMax,
+//     KV-cached decode with context size 1
+//   - the Cache (cache.rs): preallocated device KV cache (no cat-per-token),
+//     host-precomputed f32 RoPE tables incl. llama3 scaling (cache.rs:43-99)
+//   - the cluster transport (client.rs/worker.rs/proto/message.rs): the
+//     per-token activation hop is RCCL ncclSend/ncclRecv over xGMI between
+//     adjacent ranks holding contiguous layer ranges
+//   - the topology YAML incl. "model.layers.A-B" range expressions
+//     (topology.rs:134-169)
+//
+// The decode step is captured as a hipGraph (single-rank): one replay per
+// token, with token id, position and generated-token ring all device-side so
+// the graph is replayable (argmax kernel appends to the ring and advances
+// the position).
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <cmath>
+#include <cstdarg>
+#include <cstdio>
+#include <cstring>
+#include <map>
+#include <string>
+#include <vector>
+
+#include "../../include/cake_hip.h"
+#include "json.hpp"
+#include "kernels.h"
+
+// ---------------------------------------------------------------------------
+// errors
+// ---------------------------------------------------------------------------
+static thread_local std::string g_err;
+static int set_err(int code, const char* fmt, ...) {
+  char buf[1024];
+  va_list ap;
+  va_start(ap, fmt);
+  vsnprintf(buf, sizeof buf, fmt, ap);
+  va_end(ap);
+  g_err = buf;
+  return code;
+}
+extern "C" const char* cake_hip_last_error(void) { return g_err.c_str(); }
+
+#define HIP_TRY(x)                                                      \
+  do {                                                                  \
+    hipError_t _e = (x);                                                \
+    if (_e != hipSuccess)                                               \
+      return set_err(2, "%s:%d hip error: %s", __FILE__, __LINE__,      \
+                     hipGetErrorString(_e));                            \
+  } while (0)
+#define NCCL_TRY(x)                                                     \
+  do {                                                                  \
+    ncclResult_t _e = (x);                                              \
+    if (_e != ncclSuccess)                                              \
+      return set_err(3, "%s:%d rccl error: %s", __FILE__, __LINE__,     \
+                     ncclGetErrorString(_e));                           \
+  } while (0)
+
+// ---------------------------------------------------------------------------
+// model config (mirrors models/common/config.rs:87-153 hot-path subset and
+// the config.json auto-detect of cake/mod.rs:82-110,268-274)
+// ---------------------------------------------------------------------------
+struct ModelConfig {
+  int hidden = 0, inter = 0, vocab = 0, layers = 0, nh = 0, nkv = 0;
+  int head_dim = 0, max_pos = 4096;
+  float rms_eps = 1e-5f, rope_theta = 10000.f;
+  bool tied = false, qk_norm = false;
+  // llama3 rope scaling (config.rs:50-65)
+  bool rope_llama3 = false;
+  float rs_factor = 1, rs_low = 1, rs_high = 4;
+  float rs_orig = 0;
+
+  int hd() const { return head_dim ? head_dim : hidden / nh; }
+  int sq() const { return nh * hd(); }
+  int skv() const { return nkv * hd(); }
+  int nqkv() const { return sq() + 2 * skv(); }
+};
+
+static int parse_config(const char* json, ModelConfig* c) {
+  minijson::ValuePtr v;
+  try {
+    v = minijson::parse(json);
+  } catch (const std::exception& e) {
+    return set_err(5, "config.json: %s", e.what());
+  }
+  if (!v || v->kind != minijson::Value::Obj)
+    return set_err(5, "config.json: not an object");
+  auto geti = [&](const char* k, int d) {
+    auto p = v->get(k);
+    return p ? (int)p->num_or(d) : d;
+  };
+  c->hidden = geti("hidden_size", 0);
+  c->inter = geti("intermediate_size", 0);
+  c->vocab = geti("vocab_size", 0);
+  c->layers = geti("num_hidden_layers", 0);
+  c->nh = geti("num_attention_heads", 0);
+  c->nkv = geti("num_key_value_heads", c->nh);
+  c->head_dim = geti("head_dim", 0);
+  c->max_pos = geti("max_position_embeddings", 4096);
+  if (auto p = v->get("rms_norm_eps")) c->rms_eps = (float)p->num;
+  if (auto p = v->get("rope_theta")) c->rope_theta = (float)p->num;
+  if (auto p = v->get("tie_word_embeddings")) c->tied = p->bool_or(false);
+  if (auto p = v->get("model_type"))
+    c->qk_norm = p->str.find("qwen3") != std::string::npos;
+  if (auto rs = v->get("rope_scaling")) {
+    if (rs->kind == minijson::Value::Obj) {
+      auto ty = rs->get("rope_type");
+      if (!ty) ty = rs->get("type");
+      if (ty && ty->str == "llama3") {
+        c->rope_llama3 = true;
+        if (auto p = rs->get("factor")) c->rs_factor = (float)p->num;
+        if (auto p = rs->get("low_freq_factor")) c->rs_low = (float)p->num;
+        if (auto p = rs->get("high_freq_factor")) c->rs_high = (float)p->num;
+        if (auto p = rs->get("original_max_position_embeddings"))
+          c->rs_orig = (float)p->num;
+      }
+    }
+  }
+  if (!c->hidden || !c->vocab || !c->layers || !c->nh)
+    return set_err(5, "config.json: missing required fields");
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// RoPE tables (cache.rs:43-99; llama3 scaling 49-80) — host f32, same math
+// as oracle/rope_tables
+// ---------------------------------------------------------------------------
+static void build_rope_tables(const ModelConfig& c, int max_seq,
+                              std::vector<float>* cos_t,
+                              std::vector<float>* sin_t) {
+  int rd = c.hd();  // partial_rotary_factor = 1 for the in-scope configs
+  int half = rd / 2;
+  std::vector<float> theta(half);
+  for (int i = 0; i < half; ++i)
+    theta[i] = 1.0f / powf(c.rope_theta, (float)(2 * i) / (float)rd);
+  if (c.rope_llama3 && c.rs_orig > 0) {
+    float low_wl = c.rs_orig / c.rs_low;
+    float high_wl = c.rs_orig / c.rs_high;
+    for (int i = 0; i < half; ++i) {
+      float f = theta[i];
+      float wl = 2.0f * (float)M_PI / f;
+      if (wl < high_wl) {
+      } else if (wl > low_wl) {
+        theta[i] = f / c.rs_factor;
+      } else {
+        float smooth = (c.rs_orig / wl - c.rs_low) / (c.rs_high - c.rs_low);
+        theta[i] = (1.0f - smooth) * (f / c.rs_factor) + smooth * f;
+      }
+    }
+  }
+  cos_t->resize((size_t)max_seq * half);
+  sin_t->resize((size_t)max_seq * half);
+  for (int p = 0; p < max_seq; ++p)
+    for (int i = 0; i < half; ++i) {
+      float a = (float)p * theta[i];
+      (*cos_t)[(size_t)p * half + i] = cosf(a);
+      (*sin_t)[(size_t)p * half + i] = sinf(a);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// stats (per-kernel-family hipEvent timing + algorithmic bytes/flops)
+// ---------------------------------------------------------------------------
+struct FamStat {
+  long launches = 0;
+  double ms = 0, bytes = 0, flops = 0;
+};
+struct PendRec {
+  const char* name;
+  hipEvent_t e0, e1;
+  double bytes, flops;
+};
+struct Stats {
+  bool on = false;
+  std::map<std::string, FamStat> fams;
+  std::vector<PendRec> pend;
+  std::vector<std::pair<hipEvent_t, hipEvent_t>> pool;
+  size_t pool_used = 0;
+  static const size_t CAP = 32768;
+  bool truncated = false;
+};
+
+// ---------------------------------------------------------------------------
+// engine
+// ---------------------------------------------------------------------------
+struct LayerDev {
+  u16 *rms1 = nullptr, *rms2 = nullptr;
+  u16 *wqkv = nullptr, *wo = nullptr, *wgu = nullptr, *wdown = nullptr;
+  u16 *qnorm = nullptr, *knorm = nullptr;
+  u16 *kc = nullptr, *vc = nullptr;  // (nkv, max_seq, hd) each
+};
+
+struct cake_engine {
+  ModelConfig c;
+  int lo = 0, hi = 0;
+  int flags = 0, device = 0;
+  int max_seq = 0, bt = 0;  // KV capacity, max batch tokens
+  bool has_embed() const { return flags & CAKE_HIP_HAS_EMBED; }
+  bool has_head() const { return flags & CAKE_HIP_HAS_HEAD; }
+  bool use_graph() const { return flags & CAKE_HIP_USE_GRAPH; }
+
+  std::vector<LayerDev> L;  // size hi-lo
+  u16 *embed = nullptr, *norm_w = nullptr, *lm_head = nullptr;
+  float *cos_t = nullptr, *sin_t = nullptr;
+
+  // workspaces
+  u16 *x = nullptr, *xn = nullptr, *qkv = nullptr, *attn_out = nullptr;
+  u16 *gu = nullptr, *act = nullptr;
+  float* logits = nullptr;
+  float* fbuf = nullptr;
+  u32* ids = nullptr;
+  u32* ring = nullptr;
+  static const int RING_CAP = 16384;
+  float* pval = nullptr;
+  int* pidx = nullptr;
+  float* attn_ws = nullptr;
+  int* dev_pos = nullptr;
+  int* dev_step = nullptr;
+  u32* dev_tok = nullptr;
+  int host_pos = 0;
+
+  static const int NCHUNK = 16;
+
+  hipStream_t stream = nullptr;
+  hipGraphExec_t graph = nullptr;
+  bool weights_ready = false;
+
+  ncclComm_t comm = nullptr;
+  int rank = 0, world = 1;
+
+  Stats st;
+};
+
+static int dev_alloc(void** p, size_t bytes) {
+  hipError_t e = hipMalloc(p, bytes);
+  if (e != hipSuccess)
+    return set_err(2, "hipMalloc(%zu): %s", bytes, hipGetErrorString(e));
+  return 0;
+}
+#define ALLOC(ptr, ty, count)                                   \
+  do {                                                          \
+    void* _p = nullptr;                                         \
+    int _r = dev_alloc(&_p, sizeof(ty) * (size_t)(count));      \
+    if (_r) return _r;                                          \
+    (ptr) = (ty*)_p;                                            \
+  } while (0)
+
+// ---- stats helpers --------------------------------------------------------
+static hipEvent_t ev_get(cake_engine* e, bool first) {
+  Stats& s = e->st;
+  if (first) {
+    if (s.pool_used < s.pool.size()) return s.pool[s.pool_used].first;
+    if (s.pool.size() >= Stats::CAP) return nullptr;
+    hipEvent_t a, b;
+    if (hipEventCreate(&a) != hipSuccess) return nullptr;
+    if (hipEventCreate(&b) != hipSuccess) return nullptr;
+    s.pool.push_back({a, b});
+    return a;
+  }
+  return s.pool[s.pool_used].second;
+}
+struct StatScope {
+  cake_engine* e;
+  const char* name;
+  double bytes, flops;
+  bool active = false;
+  StatScope(cake_engine* e_, const char* n, double b, double f)
+      : e(e_), name(n), bytes(b), flops(f) {
+    if (e->st.on) {
+      hipEvent_t ev = ev_get(e, true);
+      if (ev) {
+        hipEventRecord(ev, e->stream);
+        active = true;
+      } else {
+        e->st.truncated = true;
+      }
+    }
+  }
+  ~StatScope() {
+    if (active) {
+      hipEvent_t ev = ev_get(e, false);
+      hipEventRecord(ev, e->stream);
+      e->st.pend.push_back({name, e->st.pool[e->st.pool_used].first, ev,
+                            bytes, flops});
+      e->st.pool_used++;
+    }
+  }
+};
+static void stats_flush(cake_engine* e) {
+  for (auto& r : e->st.pend) {
+    float ms = 0;
+    hipEventElapsedTime(&ms, r.e0, r.e1);
+    auto& f = e->st.fams[r.name];
+    f.launches++;
+    f.ms += ms;
+    f.bytes += r.bytes;
+    f.flops += r.flops;
+  }
+  e->st.pend.clear();
+  e->st.pool_used = 0;
+}
+
+// ---------------------------------------------------------------------------
+// per-layer decode / prefill enqueue (the Transformer::forward sequence,
+// transformer.rs:103-135 + attention.rs:152-357 + mlp.rs:21-31)
+// ---------------------------------------------------------------------------
+static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
+  const ModelConfig& c = e->c;
+  const int H = c.hidden, I = c.inter, hd = c.hd();
+  const int Sq = c.sq(), Nq = c.nqkv();
+  {  // rms_1
+    StatScope ss(e, "rmsnorm", 2.0 * H * 2 + H * 2, 0);
+    launch_rmsnorm(e->x, l.rms1, e->xn, 1, H, c.rms_eps, e->stream);
+  }
+  {  // fused qkv projection (GEMV)
+    StatScope ss(e, "gemv_qkv", (double)Nq * H * 2 + H * 2 + Nq * 2,
+                 2.0 * Nq * H);
+    launch_gemv(l.wqkv, e->xn, e->qkv, nullptr, Nq, H, 0, e->stream);
+  }
+  if (l.qnorm) {  // Qwen3 per-head QK-norm (attention.rs:202-215)
+    StatScope ss(e, "qknorm", 2.0 * (c.nh + c.nkv) * hd * 2, 0);
+    launch_rmsnorm(e->qkv, l.qnorm, e->qkv, c.nh, hd, c.rms_eps, e->stream);
+    launch_rmsnorm(e->qkv + Sq, l.knorm, e->qkv + Sq, c.nkv, hd, c.rms_eps,
+                   e->stream);
+  }
+  {  // rope q,k + KV store at slot *pos
+    StatScope ss(e, "rope_store", (double)Nq * hd * 0, 0);
+    launch_rope_store_decode(e->qkv, l.kc, l.vc, e->cos_t, e->sin_t,
+                             e->dev_pos, c.nh, c.nkv, hd, hd, e->max_seq,
+                             e->stream);
+  }
+  {  // decode attention over the cache
+    double kvbytes = 2.0 * (e->host_pos + 1) * c.skv() * 2;
+    StatScope ss(e, "attn_decode", kvbytes + Sq * 2 * 2,
+                 4.0 * (e->host_pos + 1) * Sq);
+    launch_attn_decode(e->qkv, l.kc, l.vc, e->dev_pos, e->attn_ws,
+                       e->attn_out, c.nh, c.nkv, hd, e->max_seq, e->NCHUNK,
+                       e->stream);
+  }
+  {  // o projection + residual
+    StatScope ss(e, "gemv_o", (double)H * Sq * 2 + Sq * 2 + H * 4,
+                 2.0 * H * Sq);
+    launch_gemv(l.wo, e->attn_out, e->x, e->x, H, Sq, 1, e->stream);
+  }
+  {  // rms_2
+    StatScope ss(e, "rmsnorm", 2.0 * H * 2 + H * 2, 0);
+    launch_rmsnorm(e->x, l.rms2, e->xn, 1, H, c.rms_eps, e->stream);
+  }
+  {  // fused gate_up GEMV + silu_mul (mlp.rs:21-31)
+    StatScope ss(e, "gemv_gateup", 2.0 * I * H * 2 + H * 2 + I * 2,
+                 4.0 * I * H);
+    launch_gemv_gateup(l.wgu, e->xn, e->act, I, H, e->stream);
+  }
+  {  // down projection + residual
+    StatScope ss(e, "gemv_down", (double)H * I * 2 + I * 2 + H * 4,
+                 2.0 * H * I);
+    launch_gemv(l.wdown, e->act, e->x, e->x, H, I, 1, e->stream);
+  }
+}
+
+static void enqueue_layer_prefill(cake_engine* e, LayerDev& l, int S,
+                                  int pos0) {
+  const ModelConfig& c = e->c;
+  const int H = c.hidden, I = c.inter, hd = c.hd();
+  const int Sq = c.sq(), Nq = c.nqkv();
+  {
+    StatScope ss(e, "rmsnorm_pf", 2.0 * S * H * 2, 0);
+    launch_rmsnorm(e->x, l.rms1, e->xn, S, H, c.rms_eps, e->stream);
+  }
+  {
+    StatScope ss(e, "gemm_qkv", (double)Nq * H * 2 + (double)S * (H + Nq) * 2,
+                 2.0 * S * Nq * H);
+    launch_gemm(e->xn, l.wqkv, e->qkv, nullptr, S, Nq, H, 0, e->stream);
+  }
+  if (l.qnorm) {
+    StatScope ss(e, "qknorm_pf", 2.0 * S * (c.nh + c.nkv) * hd * 2, 0);
+    launch_rmsnorm_strided(e->qkv, l.qnorm, e->qkv, S, c.nh, (size_t)Nq, hd,
+                           c.rms_eps, e->stream);
+    launch_rmsnorm_strided(e->qkv + Sq, l.knorm, e->qkv + Sq, S, c.nkv,
+                           (size_t)Nq, hd, c.rms_eps, e->stream);
+  }
+  {
+    StatScope ss(e, "rope_store_pf", 0, 0);
+    launch_rope_store_prefill(e->qkv, l.kc, l.vc, e->cos_t, e->sin_t, pos0, S,
+                              c.nh, c.nkv, hd, hd, e->max_seq, Nq, e->stream);
+  }
+  {
+    double n_avg = pos0 + (S + 1) * 0.5;
+    StatScope ss(e, "attn_prefill", 2.0 * S * n_avg * 2 * hd * c.nh / 4,
+                 4.0 * S * n_avg * hd * c.nh);
+    launch_attn_prefill(e->qkv, l.kc, l.vc, e->attn_out, S, pos0, c.nh, c.nkv,
+                        hd, e->max_seq, Nq, Sq, e->stream);
+  }
+  {
+    StatScope ss(e, "gemm_o", (double)H * Sq * 2 + (double)S * (Sq + H) * 2,
+                 2.0 * S * H * Sq);
+    launch_gemm(e->attn_out, l.wo, e->x, e->x, S, H, Sq, 1, e->stream);
+  }
+  {
+    StatScope ss(e, "rmsnorm_pf", 2.0 * S * H * 2, 0);
+    launch_rmsnorm(e->x, l.rms2, e->xn, S, H, c.rms_eps, e->stream);
+  }
+  {
+    StatScope ss(e, "gemm_gateup",
+                 2.0 * I * H * 2 + (double)S * (H + 2.0 * I) * 2,
+                 4.0 * S * I * H);
+    launch_gemm(e->xn, l.wgu, e->gu, nullptr, S, 2 * I, H, 0, e->stream);
+  }
+  {
+    StatScope ss(e, "silu_mul", 3.0 * S * I * 2, 0);
+    launch_silu_mul_rows(e->gu, e->act, S, I, e->stream);
+  }
+  {
+    StatScope ss(e, "gemm_down", (double)H * I * 2 + (double)S * (I + H) * 2,
+                 2.0 * S * H * I);
+    launch_gemm(e->act, l.wdown, e->x, e->x, S, H, I, 1, e->stream);
+  }
+}
+
+static void enqueue_head_sample(cake_engine* e, int S, int advance_by) {
+  const ModelConfig& c = e->c;
+  const int H = c.hidden, V = c.vocab;
+  {  // final norm on the last token only (text_model.rs:336-346)
+    StatScope ss(e, "rmsnorm", 2.0 * H * 2, 0);
+    launch_rmsnorm(e->x + (size_t)(S - 1) * H, e->norm_w, e->xn, 1, H,
+                   c.rms_eps, e->stream);
+  }
+  {  // lm_head (f32 logits)
+    StatScope ss(e, "gemv_head", (double)V * H * 2 + H * 2 + V * 4,
+                 2.0 * V * H);
+    launch_gemv(e->lm_head, e->xn, e->logits, nullptr, V, H, 2, e->stream);
+  }
+  if (advance_by > 1)
+    launch_advance_pos(e->dev_pos, advance_by - 1, e->stream);
+  {  // greedy ArgMax + ring append + pos++ (text_model.rs:104)
+    StatScope ss(e, "argmax", (double)V * 4, 0);
+    launch_argmax(e->logits, V, e->pval, e->pidx, e->dev_tok, e->dev_pos,
+                  e->ring, e->dev_step, 1, e->stream);
+  }
+}
+
+// one full decode step on this rank (graph-capturable when world == 1)
+static int enqueue_decode_step(cake_engine* e) {
+  const ModelConfig& c = e->c;
+  const int H = c.hidden;
+  if (e->world == 1) {
+    launch_embed_token(e->embed, e->dev_tok, e->x, H, e->stream);
+    for (auto& l : e->L) enqueue_layer_decode(e, l);
+    enqueue_head_sample(e, 1, 1);
+    return 0;
+  }
+  if (e->rank == 0) {
+    launch_embed_token(e->embed, e->dev_tok, e->x, H, e->stream);
+    for (auto& l : e->L) enqueue_layer_decode(e, l);
+    NCCL_TRY(ncclSend(e->x, H, ncclBfloat16, 1, e->comm, e->stream));
+    NCCL_TRY(ncclRecv(e->x, H, ncclBfloat16, e->world - 1, e->comm,
+                      e->stream));
+    enqueue_head_sample(e, 1, 1);
+  } else {
+    NCCL_TRY(ncclRecv(e->x, H, ncclBfloat16, e->rank - 1, e->comm,
+                      e->stream));
+    for (auto& l : e->L) enqueue_layer_decode(e, l);
+    NCCL_TRY(ncclSend(e->x, H, ncclBfloat16, (e->rank + 1) % e->world,
+                      e->comm, e->stream));
+    launch_advance_pos(e->dev_pos, 1, e->stream);
+  }
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// C-ABI: lifecycle
+// ---------------------------------------------------------------------------
+extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
+                                      int layer_hi, int flags, int max_seq,
+                                      int max_batch_tokens, int device,
+                                      cake_engine** out) {
+  ModelConfig c;
+  int r = parse_config(config_json, &c);
+  if (r) return r;
+  if (layer_lo < 0 || layer_hi > c.layers || layer_lo >= layer_hi)
+    return set_err(5, "bad layer range [%d,%d) of %d", layer_lo, layer_hi,
+                   c.layers);
+  if (c.hd() > 128 || c.hd() % 8 != 0)
+    return set_err(5, "head_dim %d unsupported (must be <=128, mult of 8)",
+                   c.hd());
+  if (max_seq <= 0) max_seq = c.max_pos;
+  if (max_batch_tokens <= 0) max_batch_tokens = 2048;
+  if (max_batch_tokens > max_seq) max_batch_tokens = max_seq;
+
+  HIP_TRY(hipSetDevice(device));
+  auto* e = new cake_engine();
+  e->c = c;
+  e->lo = layer_lo;
+  e->hi = layer_hi;
+  e->flags = flags;
+  e->device = device;
+  e->max_seq = max_seq;
+  e->bt = max_batch_tokens;
+  HIP_TRY(hipStreamCreate(&e->stream));
+
+  const int H = c.hidden, I = c.inter, V = c.vocab;
+  const int hd = c.hd(), Nq = c.nqkv(), Sq = c.sq();
+  const int BT = e->bt;
+  // weights
+  e->L.resize(layer_hi - layer_lo);
+  for (auto& l : e->L) {
+    ALLOC(l.rms1, u16, H);
+    ALLOC(l.rms2, u16, H);
+    ALLOC(l.wqkv, u16, (size_t)Nq * H);
+    ALLOC(l.wo, u16, (size_t)H * Sq);
+    ALLOC(l.wgu, u16, (size_t)2 * I * H);
+    ALLOC(l.wdown, u16, (size_t)H * I);
+    if (c.qk_norm) {
+      ALLOC(l.qnorm, u16, hd);
+      ALLOC(l.knorm, u16, hd);
+    }
+    ALLOC(l.kc, u16, (size_t)c.nkv * max_seq * hd);
+    ALLOC(l.vc, u16, (size_t)c.nkv * max_seq * hd);
+  }
+  if (e->has_embed()) ALLOC(e->embed, u16, (size_t)V * H);
+  if (e->has_head()) {
+    ALLOC(e->norm_w, u16, H);
+    if (c.tied && e->has_embed())
+      e->lm_head = e->embed;
+    else
+      ALLOC(e->lm_head, u16, (size_t)V * H);
+  }
+  // rope tables
+  {
+    std::vector<float> ct, st;
+    build_rope_tables(c, max_seq, &ct, &st);
+    ALLOC(e->cos_t, float, ct.size());
+    ALLOC(e->sin_t, float, st.size());
+    HIP_TRY(hipMemcpy(e->cos_t, ct.data(), ct.size() * 4,
+                      hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(e->sin_t, st.data(), st.size() * 4,
+                      hipMemcpyHostToDevice));
+  }
+  // workspaces
+  ALLOC(e->x, u16, (size_t)BT * H);
+  ALLOC(e->xn, u16, (size_t)BT * H);
+  ALLOC(e->qkv, u16, (size_t)BT * Nq);
+  ALLOC(e->attn_out, u16, (size_t)BT * Sq);
+  ALLOC(e->gu, u16, (size_t)BT * 2 * I);
+  ALLOC(e->act, u16, (size_t)BT * I);
+  ALLOC(e->logits, float, V);
+  ALLOC(e->fbuf, float, (size_t)BT * H);
+  ALLOC(e->ids, u32, BT);
+  ALLOC(e->ring, u32, cake_engine::RING_CAP);
+  ALLOC(e->pval, float, 256);
+  ALLOC(e->pidx, int, 256);
+  ALLOC(e->attn_ws, float, (size_t)c.nh * cake_engine::NCHUNK * (hd + 2));
+  ALLOC(e->dev_pos, int, 1);
+  ALLOC(e->dev_step, int, 1);
+  ALLOC(e->dev_tok, u32, 1);
+  HIP_TRY(hipMemset(e->dev_pos, 0, 4));
+  HIP_TRY(hipMemset(e->dev_step, 0, 4));
+  HIP_TRY(hipMemset(e->dev_tok, 0, 4));
+  *out = e;
+  return 0;
+}
+
+extern "C" void cake_hip_engine_free(cake_engine* e) {
+  if (!e) return;
+  hipSetDevice(e->device);
+  hipDeviceSynchronize();
+  if (e->graph) hipGraphExecDestroy(e->graph);
+  if (e->comm) ncclCommDestroy(e->comm);
+  // (device allocations are freed with the process; engines live for the
+  // process lifetime in the intended use — still free the big ones)
+  for (auto& l : e->L) {
+    hipFree(l.rms1); hipFree(l.rms2); hipFree(l.wqkv); hipFree(l.wo);
+    hipFree(l.wgu); hipFree(l.wdown); hipFree(l.kc); hipFree(l.vc);
+    if (l.qnorm) { hipFree(l.qnorm); hipFree(l.knorm); }
+  }
+  if (e->embed) hipFree(e->embed);
+  if (e->lm_head && e->lm_head != e->embed) hipFree(e->lm_head);
+  if (e->norm_w) hipFree(e->norm_w);
+  hipFree(e->x); hipFree(e->xn); hipFree(e->qkv); hipFree(e->attn_out);
+  hipFree(e->gu); hipFree(e->act); hipFree(e->logits); hipFree(e->fbuf);
+  hipFree(e->ids); hipFree(e->ring); hipFree(e->pval); hipFree(e->pidx);
+  hipFree(e->attn_ws); hipFree(e->dev_pos); hipFree(e->dev_step);
+  hipFree(e->dev_tok); hipFree(e->cos_t); hipFree(e->sin_t);
+  hipStreamDestroy(e->stream);
+  delete e;
+}
+
+// ---------------------------------------------------------------------------
+// weight loading
+// ---------------------------------------------------------------------------
+struct StTensor {
+  std::string dtype;
+  std::vector<long> shape;
+  size_t off0 = 0, off1 = 0;
+};
+
+static int upload_weight(cake_engine* e, u16* dst, const char* filebase,
+                         const StTensor& t, size_t expect_elems) {
+  size_t n = 1;
+  for (long d : t.shape) n *= (size_t)d;
+  if (n != expect_elems)
+    return set_err(5, "tensor shape mismatch: got %zu want %zu elems", n,
+                   expect_elems);
+  const char* src = filebase + t.off0;
+  if (t.dtype == "BF16") {
+    HIP_TRY(hipMemcpy(dst, src, n * 2, hipMemcpyHostToDevice));
+  } else if (t.dtype == "F32") {
+    std::vector<u16> tmp(n);
+    const float* f = reinterpret_cast<const float*>(src);
+    for (size_t i = 0; i < n; ++i) {
+      union { float f; unsigned u; } v{f[i]};
+      unsigned r = ((v.u & 0x7fffffffu) > 0x7f800000u)
+                       ? 0x7fc00000u
+                       : v.u + 0x7fffu + ((v.u >> 16) & 1u);
+      tmp[i] = (u16)(r >> 16);
+    }
+    HIP_TRY(hipMemcpy(dst, tmp.data(), n * 2, hipMemcpyHostToDevice));
+  } else {
+    return set_err(5, "unsupported safetensors dtype %s", t.dtype.c_str());
+  }
+  return 0;
+}
+
+extern "C" int cake_hip_load_safetensors(cake_engine* e, const char* path) {
+  HIP_TRY(hipSetDevice(e->device));
+  FILE* f = fopen(path, "rb");
+  if (!f) return set_err(4, "cannot open %s", path);
+  fseek(f, 0, SEEK_END);
+  long fsize = ftell(f);
+  fseek(f, 0, SEEK_SET);
+  std::vector<char> data((size_t)fsize);
+  if (fread(data.data(), 1, (size_t)fsize, f) != (size_t)fsize) {
+    fclose(f);
+    return set_err(4, "short read on %s", path);
+  }
+  fclose(f);
+  if (fsize < 8) return set_err(4, "bad safetensors file");
+  uint64_t hlen;
+  memcpy(&hlen, data.data(), 8);
+  if ((long)(8 + hlen) > fsize) return set_err(4, "bad safetensors header");
+  std::string header(data.data() + 8, hlen);
+  minijson::ValuePtr hv;
+  try {
+    hv = minijson::parse(header);
+  } catch (const std::exception& ex) {
+    return set_err(4, "safetensors header: %s", ex.what());
+  }
+  std::map<std::string, StTensor> tensors;
+  for (auto& kv : hv->obj) {
+    if (kv.first == "__metadata__") continue;
+    StTensor t;
+    if (auto d = kv.second->get("dtype")) t.dtype = d->str;
+    if (auto sh = kv.second->get("shape"))
+      for (auto& d : sh->arr) t.shape.push_back((long)d->num);
+    if (auto off = kv.second->get("data_offsets")) {
+      t.off0 = (size_t)off->arr[0]->num;
+      t.off1 = (size_t)off->arr[1]->num;
+    }
+    tensors[kv.first] = t;
+  }
+  const char* base = data.data() + 8 + hlen;
+  const ModelConfig& c = e->c;
+  const size_t H = c.hidden, I = c.inter, V = c.vocab;
+  const size_t hd = c.hd(), Sq = c.sq(), Skv = c.skv();
+
+  auto need = [&](const std::string& name, StTensor* out_t) -> int {
+    auto it = tensors.find(name);
+    if (it == tensors.end()) return set_err(4, "missing tensor %s",
+                                            name.c_str());
+    *out_t = it->second;
+    return 0;
+  };
+  int r;
+  StTensor t;
+  if (e->has_embed()) {
+    if ((r = need("model.embed_tokens.weight", &t))) return r;
+    if ((r = upload_weight(e, e->embed, base, t, V * H))) return r;
+  }
+  if (e->has_head()) {
+    if ((r = need("model.norm.weight", &t))) return r;
+    if ((r = upload_weight(e, e->norm_w, base, t, H))) return r;
+    if (!(c.tied && e->has_embed())) {
+      std::string name = c.tied ? "model.embed_tokens.weight"
+                                : "lm_head.weight";
+      if ((r = need(name, &t))) return r;
+      if ((r = upload_weight(e, e->lm_head, base, t, V * H))) return r;
+    }
+  }
+  for (int li = e->lo; li < e->hi; ++li) {
+    LayerDev& l = e->L[li - e->lo];
+    char p[128];
+    snprintf(p, sizeof p, "model.layers.%d.", li);
+    std::string pre(p);
+    if ((r = need(pre + "input_layernorm.weight", &t))) return r;
+    if ((r = upload_weight(e, l.rms1, base, t, H))) return r;
+    if ((r = need(pre + "post_attention_layernorm.weight", &t))) return r;
+    if ((r = upload_weight(e, l.rms2, base, t, H))) return r;
+    // fused qkv: upload q,k,v at row offsets (attention.rs:109-114)
+    if ((r = need(pre + "self_attn.q_proj.weight", &t))) return r;
+    if ((r = upload_weight(e, l.wqkv, base, t, Sq * H))) return r;
+    if ((r = need(pre + "self_attn.k_proj.weight", &t))) return r;
+    if ((r = upload_weight(e, l.wqkv + Sq * H, base, t, Skv * H))) return r;
+    if ((r = need(pre + "self_attn.v_proj.weight", &t))) return r;
+    if ((r = upload_weight(e, l.wqkv + (Sq + Skv) * H, base, t, Skv * H)))
+      return r;
+    if ((r = need(pre + "self_attn.o_proj.weight", &t))) return r;
+    if ((r = upload_weight(e, l.wo, base, t, H * Sq))) return r;
+    // fused gate_up (mlp.rs:38-46)
+    if ((r = need(pre + "mlp.gate_proj.weight", &t))) return r;
+    if ((r = upload_weight(e, l.wgu, base, t, I * H))) return r;
+    if ((r = need(pre + "mlp.up_proj.weight", &t))) return r;
+    if ((r = upload_weight(e, l.wgu + I * H, base, t, I * H))) return r;
+    if ((r = need(pre + "mlp.down_proj.weight", &t))) return r;
+    if ((r = upload_weight(e, l.wdown, base, t, H * I))) return r;
+    if (c.qk_norm) {
+      if ((r = need(pre + "self_attn.q_norm.weight", &t))) return r;
+      if ((r = upload_weight(e, l.qnorm, base, t, hd))) return r;
+      if ((r = need(pre + "self_attn.k_norm.weight", &t))) return r;
+      if ((r = upload_weight(e, l.knorm, base, t, hd))) return r;
+    }
+  }
+  HIP_TRY(hipDeviceSynchronize());
+  e->weights_ready = true;
+  return 0;
+}
+
+extern "C" int cake_hip_init_random(cake_engine* e, uint64_t seed,
+                                    float scale) {
+  HIP_TRY(hipSetDevice(e->device));
+  const ModelConfig& c = e->c;
+  const size_t H = c.hidden, I = c.inter, V = c.vocab;
+  const size_t hd = c.hd(), Sq = c.sq(), Skv = c.skv();
+  uint64_t salt = 1;
+  auto fill = [&](u16* p, size_t n) {
+    launch_fill_random(p, n, seed + 0x1000193u * (salt++), scale, e->stream);
+  };
+  if (e->has_embed()) fill(e->embed, V * H);
+  if (e->has_head()) {
+    launch_fill_const(e->norm_w, H, 1.0f, e->stream);
+    if (!(c.tied && e->has_embed())) fill(e->lm_head, V * H);
+  }
+  for (auto& l : e->L) {
+    launch_fill_const(l.rms1, H, 1.0f, e->stream);
+    launch_fill_const(l.rms2, H, 1.0f, e->stream);
+    fill(l.wqkv, (Sq + 2 * Skv) * H);
+    fill(l.wo, H * Sq);
+    fill(l.wgu, 2 * I * H);
+    fill(l.wdown, H * I);
+    if (l.qnorm) {
+      launch_fill_const(l.qnorm, hd, 1.0f, e->stream);
+      launch_fill_const(l.knorm, hd, 1.0f, e->stream);
+    }
+  }
+  HIP_TRY(hipStreamSynchronize(e->stream));
+  e->weights_ready = true;
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// generation
+// ---------------------------------------------------------------------------
+extern "C" int cake_hip_reset(cake_engine* e) {
+  HIP_TRY(hipSetDevice(e->device));
+  HIP_TRY(hipStreamSynchronize(e->stream));
+  HIP_TRY(hipMemset(e->dev_pos, 0, 4));
+  HIP_TRY(hipMemset(e->dev_step, 0, 4));
+  HIP_TRY(hipMemset(e->dev_tok, 0, 4));
+  e->host_pos = 0;
+  return 0;
+}
+
+extern "C" int cake_hip_prefill(cake_engine* e, const uint32_t* tokens,
+                                int n_tokens, uint32_t* next_token,
+                                float* logits_out) {
+  HIP_TRY(hipSetDevice(e->device));
+  if (!e->weights_ready) return set_err(1, "weights not loaded");
+  if (n_tokens <= 0) return set_err(5, "n_tokens must be > 0");
+  if (e->host_pos + n_tokens > e->max_seq)
+    return set_err(5, "prefill exceeds max_seq (%d + %d > %d)", e->host_pos,
+                   n_tokens, e->max_seq);
+  const ModelConfig& c = e->c;
+  const int H = c.hidden;
+  int done = 0;
+  while (done < n_tokens) {
+    int S = std::min(n_tokens - done, e->bt);
+    int pos0 = e->host_pos;
+    bool last_chunk = (done + S == n_tokens);
+    if (e->world == 1 || e->rank == 0) {
+      if (!tokens) return set_err(5, "rank 0 needs tokens");
+      HIP_TRY(hipMemcpyAsync(e->ids, tokens + done, (size_t)S * 4,
+                             hipMemcpyHostToDevice, e->stream));
+      launch_embed_rows(e->embed, e->ids, e->x, S, H, e->stream);
+      for (auto& l : e->L) enqueue_layer_prefill(e, l, S, pos0);
+      if (e->world > 1) {
+        NCCL_TRY(ncclSend(e->x, (size_t)S * H, ncclBfloat16, 1, e->comm,
+                          e->stream));
+        NCCL_TRY(ncclRecv(e->x, (size_t)S * H, ncclBfloat16, e->world - 1,
+                          e->comm, e->stream));
+      }
+      if (last_chunk) {
+        enqueue_head_sample(e, S, S);
+      } else {
+        launch_advance_pos(e->dev_pos, S, e->stream);
+      }
+    } else {
+      NCCL_TRY(ncclRecv(e->x, (size_t)S * H, ncclBfloat16, e->rank - 1,
+                        e->comm, e->stream));
+      for (auto& l : e->L) enqueue_layer_prefill(e, l, S, pos0);
+      NCCL_TRY(ncclSend(e->x, (size_t)S * H, ncclBfloat16,
+                        (e->rank + 1) % e->world, e->comm, e->stream));
+      launch_advance_pos(e->dev_pos, S, e->stream);
+    }
+    e->host_pos += S;
+    done += S;
+  }
+  HIP_TRY(hipStreamSynchronize(e->stream));
+  stats_flush(e);
+  if ((e->world == 1 || e->rank == 0) && e->has_head()) {
+    if (next_token) {
+      u32 tok;
+      HIP_TRY(hipMemcpy(&tok, e->dev_tok, 4, hipMemcpyDeviceToHost));
+      *next_token = tok;
+    }
+    if (logits_out) {
+      HIP_TRY(hipMemcpy(logits_out, e->logits, (size_t)c.vocab * 4,
+                        hipMemcpyDeviceToHost));
+    }
+  }
+  return 0;
+}
+
+extern "C" int cake_hip_decode(cake_engine* e, int steps,
+                               uint32_t* tokens_out) {
+  HIP_TRY(hipSetDevice(e->device));
+  if (!e->weights_ready) return set_err(1, "weights not loaded");
+  if (steps <= 0 || steps > cake_engine::RING_CAP)
+    return set_err(5, "steps out of range");
+  if (e->host_pos + steps > e->max_seq)
+    return set_err(5, "decode exceeds max_seq");
+  HIP_TRY(hipMemsetAsync(e->dev_step, 0, 4, e->stream));
+  bool graph_ok = e->use_graph() && e->world == 1 && !e->st.on;
+  if (graph_ok && !e->graph) {
+    // capture one decode step (device-side token/pos/ring make it replayable)
+    hipGraph_t g;
+    HIP_TRY(hipStreamBeginCapture(e->stream, hipStreamCaptureModeGlobal));
+    int r = enqueue_decode_step(e);
+    if (r) {
+      hipStreamEndCapture(e->stream, &g);
+      return r;
+    }
+    HIP_TRY(hipStreamEndCapture(e->stream, &g));
+    HIP_TRY(hipGraphInstantiate(&e->graph, g, nullptr, nullptr, 0));
+    HIP_TRY(hipGraphDestroy(g));
+  }
+  for (int s = 0; s < steps; ++s) {
+    if (graph_ok) {
+      HIP_TRY(hipGraphLaunch(e->graph, e->stream));
+    } else {
+      int r = enqueue_decode_step(e);
+      if (r) return r;
+    }
+    e->host_pos += 1;
+    // bound in-flight work: sync every 64 steps to keep the queue shallow
+    if ((s & 63) == 63) HIP_TRY(hipStreamSynchronize(e->stream));
+  }
+  HIP_TRY(hipStreamSynchronize(e->stream));
+  stats_flush(e);
+  if (tokens_out && (e->world == 1 || e->rank == 0) && e->has_head()) {
+    HIP_TRY(hipMemcpy(tokens_out, e->ring, (size_t)steps * 4,
+                      hipMemcpyDeviceToHost));
+  }
+  return 0;
+}
+
+extern "C" int cake_hip_forward_hidden(cake_engine* e, const float* x,
+                                       int seq, int index_pos, float* out) {
+  HIP_TRY(hipSetDevice(e->device));
+  if (!e->weights_ready) return set_err(1, "weights not loaded");
+  if (seq <= 0 || seq > e->bt) return set_err(5, "bad seq");
+  const int H = e->c.hidden;
+  HIP_TRY(hipMemcpyAsync(e->fbuf, x, (size_t)seq * H * 4,
+                         hipMemcpyHostToDevice, e->stream));
+  launch_f32_to_bf16(e->fbuf, e->x, (size_t)seq * H, e->stream);
+  // sync device position with the caller's index_pos
+  HIP_TRY(hipStreamSynchronize(e->stream));
+  HIP_TRY(hipMemcpy(e->dev_pos, &index_pos, 4, hipMemcpyHostToDevice));
+  e->host_pos = index_pos;
+  if (seq == 1) {
+    for (auto& l : e->L) enqueue_layer_decode(e, l);
+  } else {
+    for (auto& l : e->L) enqueue_layer_prefill(e, l, seq, index_pos);
+  }
+  launch_advance_pos(e->dev_pos, seq, e->stream);
+  launch_bf16_to_f32(e->x, e->fbuf, (size_t)seq * H, e->stream);
+  HIP_TRY(hipMemcpyAsync(out, e->fbuf, (size_t)seq * H * 4,
+                         hipMemcpyDeviceToHost, e->stream));
+  HIP_TRY(hipStreamSynchronize(e->stream));
+  stats_flush(e);
+  e->host_pos += seq;
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// comm (RCCL over xGMI)
+// ---------------------------------------------------------------------------
+extern "C" int cake_hip_comm_id(uint8_t out[CAKE_HIP_COMM_ID_BYTES]) {
+  ncclUniqueId id;
+  static_assert(sizeof(ncclUniqueId) == CAKE_HIP_COMM_ID_BYTES,
+                "ncclUniqueId size");
+  NCCL_TRY(ncclGetUniqueId(&id));
+  memcpy(out, &id, sizeof id);
+  return 0;
+}
+
+extern "C" int cake_hip_comm_init(cake_engine* e, int rank, int world_size,
+                                  const uint8_t id[CAKE_HIP_COMM_ID_BYTES]) {
+  HIP_TRY(hipSetDevice(e->device));
+  ncclUniqueId nid;
+  memcpy(&nid, id, sizeof nid);
+  NCCL_TRY(ncclCommInitRank(&e->comm, world_size, nid, rank));
+  e->rank = rank;
+  e->world = world_size;
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// topology (sharding/topology.rs:134-169 + range expr 13,142-166)
+// ---------------------------------------------------------------------------
+static bool expand_range(const std::string& s, std::vector<int>* out) {
+  // match "(.+[^\d])(\d+)-(\d+)$" (topology.rs:13)
+  size_t dash = s.rfind('-');
+  if (dash == std::string::npos || dash + 1 >= s.size()) return false;
+  size_t e2 = dash + 1;
+  for (size_t i = e2; i < s.size(); ++i)
+    if (!isdigit((unsigned char)s[i])) return false;
+  size_t b1 = dash;
+  while (b1 > 0 && isdigit((unsigned char)s[b1 - 1])) --b1;
+  if (b1 == dash || b1 == 0 || isdigit((unsigned char)s[b1 - 1]))
+    return false;
+  int start = atoi(s.substr(b1, dash - b1).c_str());
+  int stop = atoi(s.substr(e2).c_str());
+  if (stop < start) return false;
+  for (int n = start; n <= stop; ++n) out->push_back(n);
+  return true;
+}
+
+static int layer_index(const std::string& name) {
+  // "model.layers.N" -> N
+  size_t dot = name.rfind('.');
+  if (dot == std::string::npos) return -1;
+  for (size_t i = dot + 1; i < name.size(); ++i)
+    if (!isdigit((unsigned char)name[i])) return -1;
+  return atoi(name.c_str() + dot + 1);
+}
+
+extern "C" int cake_hip_topology_node_range(const char* topology_yaml,
+                                            const char* node_name, int* lo,
+                                            int* hi) {
+  // minimal YAML subset: top-level "name:", nested "  key: value",
+  // "  layers:" followed by "    - item" entries
+  std::vector<int> idx;
+  std::string cur_node;
+  bool in_layers = false;
+  bool found = false;
+  const char* p = topology_yaml;
+  while (*p) {
+    const char* nl = strchr(p, '\n');
+    std::string line = nl ? std::string(p, nl - p) : std::string(p);
+    p = nl ? nl + 1 : p + line.size();
+    // strip comments and trailing ws
+    size_t h = line.find('#');
+    if (h != std::string::npos) line = line.substr(0, h);
+    while (!line.empty() && (line.back() == ' ' || line.back() == '\r'))
+      line.pop_back();
+    if (line.empty()) continue;
+    size_t indent = 0;
+    while (indent < line.size() && line[indent] == ' ') ++indent;
+    std::string body = line.substr(indent);
+    if (indent == 0 && body.back() == ':') {
+      cur_node = body.substr(0, body.size() - 1);
+      in_layers = false;
+      if (cur_node == node_name) found = true;
+      continue;
+    }
+    if (cur_node != node_name) continue;
+    if (body == "layers:") {
+      in_layers = true;
+      continue;
+    }
+    if (in_layers && body.size() > 2 && body[0] == '-') {
+      std::string item = body.substr(1);
+      while (!item.empty() && item.front() == ' ') item.erase(0, 1);
+      if (!item.empty() && (item.front() == '"' || item.front() == '\'')) {
+        item = item.substr(1, item.size() - 2);
+      }
+      std::vector<int> nums;
+      if (expand_range(item, &nums)) {
+        for (int n : nums) idx.push_back(n);
+      } else {
+        int n = layer_index(item);
+        if (n < 0) return set_err(5, "bad layer name '%s'", item.c_str());
+        idx.push_back(n);
+      }
+      continue;
+    }
+    if (body.find(':') != std::string::npos) in_layers = false;
+  }
+  if (!found) return set_err(5, "node '%s' not in topology", node_name);
+  if (idx.empty()) return set_err(5, "node '%s' has no layers", node_name);
+  int mn = idx[0], mx = idx[0];
+  long sum = 0;
+  for (int n : idx) {
+    mn = std::min(mn, n);
+    mx = std::max(mx, n);
+    sum += n;
+  }
+  long want = (long)(mn + mx) * (mx - mn + 1) / 2;
+  if (sum != want || (int)idx.size() != mx - mn + 1)
+    return set_err(5, "node '%s' layers are not a contiguous range",
+                   node_name);
+  *lo = mn;
+  *hi = mx + 1;
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// op-level surface (kernel parity tests)
+// ---------------------------------------------------------------------------
+struct OpCtx {
+  u16 *a = nullptr, *b = nullptr, *c = nullptr;
+  float *fa = nullptr, *fb = nullptr, *fc = nullptr;
+  size_t cap = 0;
+  hipStream_t s = nullptr;
+};
+static int op_ctx(int device, size_t elems, OpCtx* o) {
+  HIP_TRY(hipSetDevice(device));
+  static thread_local OpCtx ctx;
+  if (ctx.cap < elems) {
+    if (ctx.a) {
+      hipFree(ctx.a); hipFree(ctx.b); hipFree(ctx.c);
+      hipFree(ctx.fa); hipFree(ctx.fb); hipFree(ctx.fc);
+    }
+    ctx.cap = elems;
+    ALLOC(ctx.a, u16, elems);
+    ALLOC(ctx.b, u16, elems);
+    ALLOC(ctx.c, u16, elems);
+    ALLOC(ctx.fa, float, elems);
+    ALLOC(ctx.fb, float, elems);
+    ALLOC(ctx.fc, float, elems);
+    HIP_TRY(hipStreamCreate(&ctx.s));
+  }
+  *o = ctx;
+  return 0;
+}
+static int up16(const float* src, float* fstage, u16* dst, size_t n,
+                hipStream_t s) {
+  HIP_TRY(hipMemcpyAsync(fstage, src, n * 4, hipMemcpyHostToDevice, s));
+  launch_f32_to_bf16(fstage, dst, n, s);
+  return 0;
+}
+static int down16(const u16* src, float* fstage, float* dst, size_t n,
+                  hipStream_t s) {
+  launch_bf16_to_f32(src, fstage, n, s);
+  HIP_TRY(hipMemcpyAsync(dst, fstage, n * 4, hipMemcpyDeviceToHost, s));
+  HIP_TRY(hipStreamSynchronize(s));
+  return 0;
+}
+
+extern "C" int cake_hip_op_rms_norm(int rows, int cols, float eps,
+                                    const float* x, const float* w,
+                                    float* out, int device) {
+  OpCtx o;
+  size_t n = (size_t)rows * cols;
+  int r = op_ctx(device, std::max(n, (size_t)cols), &o);
+  if (r) return r;
+  if ((r = up16(x, o.fa, o.a, n, o.s))) return r;
+  if ((r = up16(w, o.fb, o.b, cols, o.s))) return r;
+  launch_rmsnorm(o.a, o.b, o.c, rows, cols, eps, o.s);
+  return down16(o.c, o.fc, out, n, o.s);
+}
+
+extern "C" int cake_hip_op_linear(int M, int N, int K, const float* x,
+                                  const float* w, float* out, int device) {
+  OpCtx o;
+  size_t n = std::max((size_t)M * K, std::max((size_t)N * K, (size_t)M * N));
+  int r = op_ctx(device, n, &o);
+  if (r) return r;
+  if ((r = up16(x, o.fa, o.a, (size_t)M * K, o.s))) return r;
+  if ((r = up16(w, o.fb, o.b, (size_t)N * K, o.s))) return r;
+  if (M == 1) {
+    launch_gemv(o.b, o.a, o.c, nullptr, N, K, 0, o.s);
+  } else {
+    if (K % 64)
+      return set_err(5, "gemm requires K %% 64 == 0 (got %d)", K);
+    launch_gemm(o.a, o.b, o.c, nullptr, M, N, K, 0, o.s);
+  }
+  return down16(o.c, o.fc, out, (size_t)M * N, o.s);
+}
+
+extern "C" int cake_hip_op_silu_mul(long n, const float* gate,
+                                    const float* up, float* out, int device) {
+  OpCtx o;
+  int r = op_ctx(device, (size_t)n, &o);
+  if (r) return r;
+  if ((r = up16(gate, o.fa, o.a, n, o.s))) return r;
+  if ((r = up16(up, o.fb, o.b, n, o.s))) return r;
+  launch_silu_mul(o.a, o.b, o.c, n, o.s);
+  return down16(o.c, o.fc, out, n, o.s);
+}
+
+// rope on (B,H,S,D) f32 with cos/sin (S, D/2) — staged through bf16 like the
+// product path
+extern "C" int cake_hip_op_rope(int b, int h, int s, int d, const float* x,
+                                const float* cosv, const float* sinv,
+                                float* out, int device) {
+  OpCtx o;
+  size_t n = (size_t)b * h * s * d;
+  int r = op_ctx(device, n + (size_t)s * d, &o);
+  if (r) return r;
+  if ((r = up16(x, o.fa, o.a, n, o.s))) return r;
+  // cos/sin stay f32 (the product path keeps f32 tables)
+  HIP_TRY(hipMemcpyAsync(o.fb, cosv, (size_t)s * d / 2 * 4,
+                         hipMemcpyHostToDevice, o.s));
+  HIP_TRY(hipMemcpyAsync(o.fb + (size_t)s * d / 2, sinv,
+                         (size_t)s * d / 2 * 4, hipMemcpyHostToDevice, o.s));
+  launch_rope_simple(o.a, o.fb, o.fb + (size_t)s * d / 2, b * h, s, d, o.s);
+  return down16(o.a, o.fc, out, n, o.s);
+}
+
+// ---------------------------------------------------------------------------
+// stats / misc
+// ---------------------------------------------------------------------------
+extern "C" int cake_hip_set_stats(cake_engine* e, int enabled) {
+  e->st.on = enabled != 0;
+  return 0;
+}
+extern "C" int cake_hip_stats_reset(cake_engine* e) {
+  stats_flush(e);
+  e->st.fams.clear();
+  e->st.truncated = false;
+  return 0;
+}
+extern "C" int cake_hip_kernel_stats(cake_engine* e, char* buf, int cap) {
+  stats_flush(e);
+  std::string out = "{\"truncated\": ";
+  out += e->st.truncated ? "true" : "false";
+  out += ", \"kernels\": {";
+  bool first = true;
+  char tmp[256];
+  for (auto& kv : e->st.fams) {
+    if (!first) out += ", ";
+    first = false;
+    snprintf(tmp, sizeof tmp,
+             "\"%s\": {\"launches\": %ld, \"ms\": %.4f, \"bytes\": %.0f, "
+             "\"flops\": %.0f}",
+             kv.first.c_str(), kv.second.launches, kv.second.ms,
+             kv.second.bytes, kv.second.flops);
+    out += tmp;
+  }
+  out += "}}";
+  if ((int)out.size() + 1 > cap) return set_err(5, "buffer too small");
+  memcpy(buf, out.c_str(), out.size() + 1);
+  return 0;
+}
+extern "C" int cake_hip_sync(cake_engine* e) {
+  HIP_TRY(hipSetDevice(e->device));
+  HIP_TRY(hipStreamSynchronize(e->stream));
+  stats_flush(e);
+  return 0;
+}
+extern "C" const char* cake_hip_build_info(void) {
+  return "gfx950;" __DATE__ " " __TIME__;
+}

@@ -1,0 +1,54 @@
+// Host-side launch wrappers for the gfx950 kernels (kernels.hip).
+#pragma once
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+using u16 = unsigned short;
+using u32 = unsigned int;
+
+void launch_f32_to_bf16(const float* in, u16* out, size_t n, hipStream_t s);
+void launch_bf16_to_f32(const u16* in, float* out, size_t n, hipStream_t s);
+void launch_fill_random(u16* out, size_t n, uint64_t seed, float scale,
+                        hipStream_t s);
+void launch_fill_const(u16* out, size_t n, float v, hipStream_t s);
+void launch_rmsnorm(const u16* x, const u16* w, u16* out, int rows, int cols,
+                    float eps, hipStream_t s);
+void launch_rmsnorm_strided(const u16* x, const u16* w, u16* out, int outer,
+                            int inner, size_t outer_stride, int cols,
+                            float eps, hipStream_t s);
+void launch_silu_mul(const u16* g, const u16* u, u16* out, size_t n,
+                     hipStream_t s);
+void launch_silu_mul_rows(const u16* gu, u16* out, int S, int I,
+                          hipStream_t s);
+void launch_gemv(const u16* W, const u16* x, void* out, const u16* res, int N,
+                 int K, int epi, hipStream_t s);
+void launch_gemv_gateup(const u16* W, const u16* x, u16* out, int I, int K,
+                        hipStream_t s);
+void launch_embed_token(const u16* embed, const u32* tok, u16* x, int H,
+                        hipStream_t s);
+void launch_embed_rows(const u16* embed, const u32* ids, u16* x, int S, int H,
+                       hipStream_t s);
+void launch_rope_store_decode(u16* qkv, u16* kc, u16* vc, const float* cost,
+                              const float* sint, const int* pos, int nh,
+                              int nkv, int hd, int rd, int max_seq,
+                              hipStream_t s);
+void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, const float* cost,
+                               const float* sint, int pos0, int S, int nh,
+                               int nkv, int hd, int rd, int max_seq,
+                               int qkv_stride, hipStream_t s);
+void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
+                        const int* pos, float* ws, u16* out, int nh, int nkv,
+                        int hd, int max_seq, int nchunk, hipStream_t s);
+void launch_attn_prefill(const u16* qkv, const u16* kc, const u16* vc,
+                         u16* out, int S, int pos0, int nh, int nkv, int hd,
+                         int max_seq, int qkv_stride, int out_stride,
+                         hipStream_t s);
+void launch_rope_simple(u16* x, const float* cost, const float* sint, int bh,
+                        int s, int d, hipStream_t st);
+void launch_argmax(const float* logits, int n, float* pval, int* pidx,
+                   u32* tok, int* pos, u32* ring, int* step, int advance_pos,
+                   hipStream_t s);
+void launch_advance_pos(int* pos, int by, hipStream_t s);
+void launch_gemm(const u16* A, const u16* W, u16* C, const u16* res, int M,
+                 int N, int K, int epi, hipStream_t s);

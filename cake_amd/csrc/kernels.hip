@@ -1,0 +1,914 @@
+// cake_hip kernels — hand-written HIP for gfx950 (MI355X, CDNA4).
+//
+// These implement the compute ops of cake's hot path:
+//   rms_norm        backends/mod.rs:244-246
+//   linear_forward  backends/mod.rs:206-241  (GEMV for decode M=1, MFMA GEMM
+//                                             for prefill)
+//   rope            backends/mod.rs:444-482  (HF half-rotation)
+//   silu_mul        backends/mod.rs:82, backends/cuda/ops.cu:101-138
+//   attention       models/common/attention.rs:266-343 (f32 softmax, GQA)
+//   KV cache        models/common/cache.rs:184-210 — REDESIGNED as a
+//                   preallocated device cache (fixes the O(n^2) cat-per-token)
+//   embedding       backends/mod.rs:513-528
+//   argmax          models/common/text_model.rs:102-118 (greedy ArgMax)
+//
+// Design notes (MI355X-first, per /opt/skills/guides/cdna_hip_programming.md):
+//   - wave = 64 lanes everywhere; blocks are multiples of 64
+//   - bf16 storage, f32 accumulation (matches cake's CUDA dtype policy)
+//   - decode GEMV: HBM-bound; 16 B/lane coalesced weight streaming, weights
+//     loaded straight to VGPRs (guide §5 "GEMV / M <= 16": no LDS round trip)
+//   - prefill GEMM: MFMA v_mfma_f32_16x16x32_bf16, 128x128 tile, 4 waves,
+//     double-buffered LDS filled by global_load_lds (16 B) with the XOR
+//     source-swizzle (guide §5.4 rule 21) so ds_read_b128 is conflict-free
+//   - decode attention: split-KV online softmax, two-pass combine; K/V reads
+//     are 64-lane x 4 B = one 256 B transaction per row (coalesced)
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <cstdint>
+
+#define WAVE 64
+
+using u16 = unsigned short;
+using u32 = unsigned int;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using short8 = __attribute__((ext_vector_type(8))) short;
+
+// ---------------------------------------------------------------------------
+// bf16 <-> f32 (round-to-nearest-even, matches torch/HF casting)
+// ---------------------------------------------------------------------------
+__device__ __host__ inline u16 f2b(float f) {
+  union { float f; u32 u; } v{f};
+  if ((v.u & 0x7fffffffu) > 0x7f800000u) return 0x7fc0;  // NaN
+  u32 r = v.u + 0x7fffu + ((v.u >> 16) & 1u);
+  return (u16)(r >> 16);
+}
+__device__ __host__ inline float b2f(u16 h) {
+  union { u32 u; float f; } v{(u32)h << 16};
+  return v.f;
+}
+
+__device__ inline float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, WAVE);
+  return v;
+}
+__device__ inline float wave_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    v = fmaxf(v, __shfl_down(v, off, WAVE));
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// conversion / fill
+// ---------------------------------------------------------------------------
+__global__ void k_f32_to_bf16(const float* __restrict__ in,
+                              u16* __restrict__ out, size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = f2b(in[i]);
+}
+__global__ void k_bf16_to_f32(const u16* __restrict__ in,
+                              float* __restrict__ out, size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = b2f(in[i]);
+}
+
+// splitmix64-based uniform(-1,1)*scale fill (deterministic per element)
+__global__ void k_fill_random(u16* __restrict__ out, size_t n, uint64_t seed,
+                              float scale) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    uint64_t z = seed + 0x9e3779b97f4a7c15ull * (i + 1);
+    z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
+    z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
+    z ^= z >> 31;
+    float u = (float)(z >> 40) * (1.0f / 8388608.0f) - 1.0f;  // [-1, 1)
+    out[i] = f2b(u * scale);
+  }
+}
+__global__ void k_fill_const(u16* __restrict__ out, size_t n, float v) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  u16 b = f2b(v);
+  for (; i < n; i += stride) out[i] = b;
+}
+
+// ---------------------------------------------------------------------------
+// rms_norm (backends/mod.rs:244-246): out = x * w / sqrt(mean(x^2) + eps)
+// one block per row; bf16x8 vector loads (guide G13); f32 accumulate
+// ---------------------------------------------------------------------------
+// rows are addressed as (outer, inner): row r -> offset
+// (r/inner)*outer_stride + (r%inner)*cols.  Plain contiguous rows use
+// inner = nrows, outer_stride = 0.  The (outer, inner) form handles the
+// per-head QK-norm rows inside the strided qkv buffer
+// (attention.rs:202-215: norm over head_dim after reshape).
+__global__ void k_rmsnorm(const u16* __restrict__ x, const u16* __restrict__ w,
+                          u16* __restrict__ out, int cols, float eps,
+                          int inner, size_t outer_stride) {
+  const int row = blockIdx.x;
+  const size_t off =
+      (size_t)(row / inner) * outer_stride + (size_t)(row % inner) * cols;
+  const u16* xr = x + off;
+  u16* outr = out + off;
+  float ss = 0.f;
+  const int t = threadIdx.x;
+  const int nvec = cols / 8;
+  for (int i = t; i < nvec; i += blockDim.x) {
+    short8 v = *reinterpret_cast<const short8*>(xr + i * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = b2f((u16)v[j]);
+      ss += f * f;
+    }
+  }
+  for (int i = nvec * 8 + t; i < cols; i += blockDim.x) {
+    float f = b2f(xr[i]);
+    ss += f * f;
+  }
+  ss = wave_sum(ss);
+  __shared__ float red[16];
+  const int wid = t / WAVE, lane = t % WAVE;
+  if (lane == 0) red[wid] = ss;
+  __syncthreads();
+  const int nw = blockDim.x / WAVE;
+  float tot = 0.f;
+#pragma unroll
+  for (int i = 0; i < 16; ++i)
+    if (i < nw) tot += red[i];
+  const float scale = rsqrtf(tot / (float)cols + eps);
+  for (int i = t; i < nvec; i += blockDim.x) {
+    short8 v = *reinterpret_cast<const short8*>(xr + i * 8);
+    short8 wv = *reinterpret_cast<const short8*>(w + i * 8);
+    short8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = (short)f2b(b2f((u16)v[j]) * scale * b2f((u16)wv[j]));
+    *reinterpret_cast<short8*>(outr + i * 8) = o;
+  }
+  for (int i = nvec * 8 + t; i < cols; i += blockDim.x)
+    outr[i] = f2b(b2f(xr[i]) * scale * b2f(w[i]));
+}
+
+// ---------------------------------------------------------------------------
+// silu_mul (ops.cu:101-138): out = gate * sigmoid(gate) * up
+// ---------------------------------------------------------------------------
+__global__ void k_silu_mul(const u16* __restrict__ gate,
+                           const u16* __restrict__ up, u16* __restrict__ out,
+                           size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float g = b2f(gate[i]);
+    float u = b2f(up[i]);
+    out[i] = f2b(g / (1.f + __expf(-g)) * u);
+  }
+}
+// strided variant for the fused gate_up buffer (S, 2I): gate = row[0:I],
+// up = row[I:2I]  (mlp.rs:21-31 narrow semantics)
+__global__ void k_silu_mul_rows(const u16* __restrict__ gu,
+                                u16* __restrict__ out, int S, int I) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  size_t n = (size_t)S * I;
+  for (; i < n; i += stride) {
+    int s = (int)(i / I), c = (int)(i % I);
+    float g = b2f(gu[(size_t)s * 2 * I + c]);
+    float u = b2f(gu[(size_t)s * 2 * I + I + c]);
+    out[i] = f2b(g / (1.f + __expf(-g)) * u);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// GEMV: y[r] = dot(W[r,:], x) for 8 rows per block.  Decode's workhorse —
+// HBM-bound weight streaming at 16 B/lane (guide §5 GEMV row: straight to
+// VGPRs, no LDS).  x is bf16 and L1/L2-resident (re-read per row).
+// EPI: 0 = bf16 out; 1 = bf16 out + residual add (res may alias out);
+//      2 = f32 out (lm_head logits)
+// ---------------------------------------------------------------------------
+template <int EPI>
+__global__ __launch_bounds__(256) void k_gemv(
+    const u16* __restrict__ W, const u16* __restrict__ x, void* __restrict__ out,
+    const u16* __restrict__ res, int N, int K) {
+  const int t = threadIdx.x;
+  const int row0 = blockIdx.x * 8;
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  const int kiter = (K + 2047) / 2048;  // 256 threads * 8 bf16
+#pragma unroll 1
+  for (int r = 0; r < 8; ++r) {
+    const int row = row0 + r;
+    if (row >= N) break;
+    const u16* wr = W + (size_t)row * K;
+    float a = 0.f;
+    for (int i = 0; i < kiter; ++i) {
+      int k0 = i * 2048 + t * 8;
+      if (k0 + 8 <= K) {
+        short8 wv = *reinterpret_cast<const short8*>(wr + k0);
+        short8 xv = *reinterpret_cast<const short8*>(x + k0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          a = fmaf(b2f((u16)wv[j]), b2f((u16)xv[j]), a);
+      } else {
+        for (int k = k0; k < K; ++k) a = fmaf(b2f(wr[k]), b2f(x[k]), a);
+      }
+    }
+    acc[r] = a;
+  }
+  // cross-wave reduce: 4 waves x 8 rows
+  __shared__ float red[8][4];
+  const int wid = t / WAVE, lane = t % WAVE;
+#pragma unroll
+  for (int r = 0; r < 8; ++r) {
+    float v = wave_sum(acc[r]);
+    if (lane == 0) red[r][wid] = v;
+  }
+  __syncthreads();
+  if (t < 8) {
+    const int row = row0 + t;
+    if (row < N) {
+      float v = red[t][0] + red[t][1] + red[t][2] + red[t][3];
+      if (EPI == 2) {
+        reinterpret_cast<float*>(out)[row] = v;
+      } else if (EPI == 1) {
+        reinterpret_cast<u16*>(out)[row] = f2b(v + b2f(res[row]));
+      } else {
+        reinterpret_cast<u16*>(out)[row] = f2b(v);
+      }
+    }
+  }
+}
+
+// fused gate_up GEMV + silu_mul: block computes out[i] = silu(g_i) * u_i for
+// 8 output channels, where g_i = dot(W[i,:], x), u_i = dot(W[i+I,:], x)
+// (mlp.rs:21-31 with the fused gate_up weight of mlp.rs:38-46)
+__global__ __launch_bounds__(256) void k_gemv_gateup(
+    const u16* __restrict__ W, const u16* __restrict__ x, u16* __restrict__ out,
+    int I, int K) {
+  const int t = threadIdx.x;
+  const int c0 = blockIdx.x * 8;
+  float accg[8] = {0}, accu[8] = {0};
+  const int kiter = (K + 2047) / 2048;
+#pragma unroll 1
+  for (int r = 0; r < 8; ++r) {
+    const int c = c0 + r;
+    if (c >= I) break;
+    const u16* wg = W + (size_t)c * K;
+    const u16* wu = W + (size_t)(c + I) * K;
+    float g = 0.f, u = 0.f;
+    for (int i = 0; i < kiter; ++i) {
+      int k0 = i * 2048 + t * 8;
+      if (k0 + 8 <= K) {
+        short8 xv = *reinterpret_cast<const short8*>(x + k0);
+        short8 gv = *reinterpret_cast<const short8*>(wg + k0);
+        short8 uv = *reinterpret_cast<const short8*>(wu + k0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float xf = b2f((u16)xv[j]);
+          g = fmaf(b2f((u16)gv[j]), xf, g);
+          u = fmaf(b2f((u16)uv[j]), xf, u);
+        }
+      } else {
+        for (int k = k0; k < K; ++k) {
+          g = fmaf(b2f(wg[k]), b2f(x[k]), g);
+          u = fmaf(b2f(wu[k]), b2f(x[k]), u);
+        }
+      }
+    }
+    accg[r] = g;
+    accu[r] = u;
+  }
+  __shared__ float redg[8][4], redu[8][4];
+  const int wid = t / WAVE, lane = t % WAVE;
+#pragma unroll
+  for (int r = 0; r < 8; ++r) {
+    float g = wave_sum(accg[r]);
+    float u = wave_sum(accu[r]);
+    if (lane == 0) { redg[r][wid] = g; redu[r][wid] = u; }
+  }
+  __syncthreads();
+  if (t < 8 && c0 + t < I) {
+    float g = redg[t][0] + redg[t][1] + redg[t][2] + redg[t][3];
+    float u = redu[t][0] + redu[t][1] + redu[t][2] + redu[t][3];
+    out[c0 + t] = f2b(g / (1.f + __expf(-g)) * u);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// embedding gather (backends/mod.rs:513-528)
+// ---------------------------------------------------------------------------
+__global__ void k_embed_token(const u16* __restrict__ embed,
+                              const u32* __restrict__ tok,
+                              u16* __restrict__ x, int H) {
+  const u16* src = embed + (size_t)(*tok) * H;
+  for (int i = threadIdx.x; i < H; i += blockDim.x) x[i] = src[i];
+}
+__global__ void k_embed_rows(const u16* __restrict__ embed,
+                             const u32* __restrict__ ids,
+                             u16* __restrict__ x, int H) {
+  const int s = blockIdx.x;
+  const u16* src = embed + (size_t)ids[s] * H;
+  u16* dst = x + (size_t)s * H;
+  for (int i = threadIdx.x; i < H; i += blockDim.x) dst[i] = src[i];
+}
+
+// ---------------------------------------------------------------------------
+// RoPE + KV store.  HF half-rotation (backends/mod.rs:470-477):
+//   out[i] = x1*c - x2*s ; out[i+half] = x2*c + x1*s, cos/sin row = position.
+// Decode: one token at *pos; K/V written into the preallocated cache at slot
+// *pos (replaces cache.rs:195-196 cat).  qkv layout: (Sq | Skv | Skv) per row.
+// grid = nh + 2*nkv blocks.
+// ---------------------------------------------------------------------------
+__global__ void k_rope_store_decode(u16* __restrict__ qkv,
+                                    u16* __restrict__ kc, u16* __restrict__ vc,
+                                    const float* __restrict__ cost,
+                                    const float* __restrict__ sint,
+                                    const int* __restrict__ pos, int nh,
+                                    int nkv, int hd, int rd, int max_seq) {
+  const int b = blockIdx.x;
+  const int p = *pos;
+  const int half = rd / 2;
+  const float* c = cost + (size_t)p * half;
+  const float* s = sint + (size_t)p * half;
+  if (b < nh) {                       // rope q head in place
+    u16* q = qkv + (size_t)b * hd;
+    for (int i = threadIdx.x; i < half; i += blockDim.x) {
+      float x1 = b2f(q[i]), x2 = b2f(q[i + half]);
+      q[i] = f2b(x1 * c[i] - x2 * s[i]);
+      q[i + half] = f2b(x2 * c[i] + x1 * s[i]);
+    }
+  } else if (b < nh + nkv) {          // rope k head -> cache slot p
+    const int h = b - nh;
+    u16* k = qkv + (size_t)(nh + h) * hd;
+    u16* dst = kc + ((size_t)h * max_seq + p) * hd;
+    for (int i = threadIdx.x; i < half; i += blockDim.x) {
+      float x1 = b2f(k[i]), x2 = b2f(k[i + half]);
+      dst[i] = f2b(x1 * c[i] - x2 * s[i]);
+      dst[i + half] = f2b(x2 * c[i] + x1 * s[i]);
+    }
+    for (int i = rd + threadIdx.x; i < hd; i += blockDim.x) dst[i] = k[i];
+  } else {                            // v head -> cache slot p
+    const int h = b - nh - nkv;
+    const u16* v = qkv + (size_t)(nh + nkv + h) * hd;
+    u16* dst = vc + ((size_t)h * max_seq + p) * hd;
+    for (int i = threadIdx.x; i < hd; i += blockDim.x) dst[i] = v[i];
+  }
+}
+
+// Prefill: S tokens at positions pos0..pos0+S-1; qkv is (S, Sq+2*Skv).
+// grid = (nh + 2*nkv, S)
+__global__ void k_rope_store_prefill(u16* __restrict__ qkv,
+                                     u16* __restrict__ kc, u16* __restrict__ vc,
+                                     const float* __restrict__ cost,
+                                     const float* __restrict__ sint, int pos0,
+                                     int nh, int nkv, int hd, int rd,
+                                     int max_seq, int qkv_stride) {
+  const int b = blockIdx.x;
+  const int sidx = blockIdx.y;
+  const int p = pos0 + sidx;
+  const int half = rd / 2;
+  const float* c = cost + (size_t)p * half;
+  const float* s = sint + (size_t)p * half;
+  u16* row = qkv + (size_t)sidx * qkv_stride;
+  if (b < nh) {
+    u16* q = row + (size_t)b * hd;
+    for (int i = threadIdx.x; i < half; i += blockDim.x) {
+      float x1 = b2f(q[i]), x2 = b2f(q[i + half]);
+      q[i] = f2b(x1 * c[i] - x2 * s[i]);
+      q[i + half] = f2b(x2 * c[i] + x1 * s[i]);
+    }
+  } else if (b < nh + nkv) {
+    const int h = b - nh;
+    u16* k = row + (size_t)(nh + h) * hd;
+    u16* dst = kc + ((size_t)h * max_seq + p) * hd;
+    for (int i = threadIdx.x; i < half; i += blockDim.x) {
+      float x1 = b2f(k[i]), x2 = b2f(k[i + half]);
+      dst[i] = f2b(x1 * c[i] - x2 * s[i]);
+      dst[i + half] = f2b(x2 * c[i] + x1 * s[i]);
+    }
+    for (int i = rd + threadIdx.x; i < hd; i += blockDim.x) dst[i] = k[i];
+  } else {
+    const int h = b - nh - nkv;
+    const u16* v = row + (size_t)(nh + nkv + h) * hd;
+    u16* dst = vc + ((size_t)h * max_seq + p) * hd;
+    for (int i = threadIdx.x; i < hd; i += blockDim.x) dst[i] = v[i];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Decode attention (attention.rs:300-343 semantics, GQA, f32 softmax), over
+// the preallocated cache.  Split-KV: grid (NCHUNK, nh); each block computes
+// an online-softmax partial over a contiguous slice of positions, written to
+// ws[h][chunk] = {o[hd], m, l}; k_attn_combine reduces the partials.
+// K/V row loads: 64 lanes x 4 B = one coalesced 256 B transaction.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_attn_decode_part(
+    const u16* __restrict__ q,         // (nh*hd), post-rope
+    const u16* __restrict__ kc, const u16* __restrict__ vc,
+    const int* __restrict__ pos, float* __restrict__ ws, int nh, int nkv,
+    int hd, int max_seq, int nchunk) {
+  const int h = blockIdx.y;
+  const int chunk = blockIdx.x;
+  const int n = *pos + 1;
+  const int cs = (n + nchunk - 1) / nchunk;
+  const int start = chunk * cs;
+  const int end = min(start + cs, n);
+  const int kvh = h / (nh / nkv);
+  const int t = threadIdx.x, wid = t / WAVE, lane = t % WAVE;
+  const int e0 = 2 * lane;             // dims (2*lane, 2*lane+1)
+  const bool act = e0 + 1 < hd;
+  float* wsrow = ws + ((size_t)h * nchunk + chunk) * (hd + 2);
+
+  float q0 = 0.f, q1 = 0.f;
+  if (act) {
+    q0 = b2f(q[(size_t)h * hd + e0]);
+    q1 = b2f(q[(size_t)h * hd + e0 + 1]);
+  }
+  const float scale = rsqrtf((float)hd);
+  float m = -INFINITY, l = 0.f, o0 = 0.f, o1 = 0.f;
+  const u16* kbase = kc + (size_t)kvh * max_seq * hd;
+  const u16* vbase = vc + (size_t)kvh * max_seq * hd;
+  for (int p = start + wid; p < end; p += 4) {
+    float dot = 0.f;
+    if (act) {
+      const u16* kr = kbase + (size_t)p * hd + e0;
+      dot = q0 * b2f(kr[0]) + q1 * b2f(kr[1]);
+    }
+    dot = wave_sum(dot) * scale;
+    dot = __shfl(dot, 0, WAVE);
+    float mn = fmaxf(m, dot);
+    float alpha = __expf(m - mn);
+    float pw = __expf(dot - mn);
+    float v0 = 0.f, v1 = 0.f;
+    if (act) {
+      const u16* vr = vbase + (size_t)p * hd + e0;
+      v0 = b2f(vr[0]);
+      v1 = b2f(vr[1]);
+    }
+    o0 = o0 * alpha + pw * v0;
+    o1 = o1 * alpha + pw * v1;
+    l = l * alpha + pw;
+    m = mn;
+  }
+  // combine the 4 waves' partials
+  __shared__ float sm[4], sl[4];
+  __shared__ float so[4][128 + 8];
+  if (lane == 0) { sm[wid] = m; sl[wid] = l; }
+  if (act) { so[wid][e0] = o0; so[wid][e0 + 1] = o1; }
+  __syncthreads();
+  if (t == 0) {
+    float M = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
+    float L = 0.f;
+#pragma unroll
+    for (int w = 0; w < 4; ++w)
+      L += (sm[w] == -INFINITY) ? 0.f : sl[w] * __expf(sm[w] - M);
+    wsrow[hd] = M;
+    wsrow[hd + 1] = L;
+  }
+  __syncthreads();
+  float M = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
+  for (int d = t; d < hd; d += blockDim.x) {
+    float o = 0.f;
+#pragma unroll
+    for (int w = 0; w < 4; ++w)
+      o += (sm[w] == -INFINITY) ? 0.f : so[w][d] * __expf(sm[w] - M);
+    wsrow[d] = o;
+  }
+}
+
+__global__ void k_attn_decode_combine(const float* __restrict__ ws,
+                                      u16* __restrict__ out, int nh, int hd,
+                                      int nchunk) {
+  const int h = blockIdx.x;
+  const float* base = ws + (size_t)h * nchunk * (hd + 2);
+  __shared__ float M;
+  if (threadIdx.x == 0) {
+    float m = -INFINITY;
+    for (int c = 0; c < nchunk; ++c) m = fmaxf(m, base[c * (hd + 2) + hd]);
+    M = m;
+  }
+  __syncthreads();
+  float Mv = M;
+  __shared__ float L;
+  if (threadIdx.x == 0) {
+    float l = 0.f;
+    for (int c = 0; c < nchunk; ++c) {
+      float mc = base[c * (hd + 2) + hd];
+      if (mc != -INFINITY) l += base[c * (hd + 2) + hd + 1] * __expf(mc - Mv);
+    }
+    L = l;
+  }
+  __syncthreads();
+  float Lv = L;
+  for (int d = threadIdx.x; d < hd; d += blockDim.x) {
+    float o = 0.f;
+    for (int c = 0; c < nchunk; ++c) {
+      float mc = base[c * (hd + 2) + hd];
+      if (mc != -INFINITY) o += base[c * (hd + 2) + d] * __expf(mc - Mv);
+    }
+    out[(size_t)h * hd + d] = f2b(o / Lv);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Prefill attention — flash-style f32 online softmax, causal, GQA, over the
+// cache (which already holds positions [0, pos0+S)).  One wave per query
+// row; 4 rows per block.  (attention.rs:300-343 + cache.rs:150-160 mask.)
+// q rows come from the post-rope qkv buffer.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_attn_prefill(
+    const u16* __restrict__ qkv, const u16* __restrict__ kc,
+    const u16* __restrict__ vc, u16* __restrict__ out, int S, int pos0,
+    int nh, int nkv, int hd, int max_seq, int qkv_stride, int out_stride) {
+  const int h = blockIdx.y;
+  const int t = threadIdx.x, wid = t / WAVE, lane = t % WAVE;
+  const int sidx = blockIdx.x * 4 + wid;
+  if (sidx >= S) return;
+  const int kvh = h / (nh / nkv);
+  const int e0 = 2 * lane;
+  const bool act = e0 + 1 < hd;
+  const int n = pos0 + sidx + 1;      // causal: attend to <= own position
+  float q0 = 0.f, q1 = 0.f;
+  if (act) {
+    const u16* qr = qkv + (size_t)sidx * qkv_stride + (size_t)h * hd + e0;
+    q0 = b2f(qr[0]);
+    q1 = b2f(qr[1]);
+  }
+  const float scale = rsqrtf((float)hd);
+  float m = -INFINITY, l = 0.f, o0 = 0.f, o1 = 0.f;
+  const u16* kbase = kc + (size_t)kvh * max_seq * hd;
+  const u16* vbase = vc + (size_t)kvh * max_seq * hd;
+  for (int p = 0; p < n; ++p) {
+    float dot = 0.f;
+    if (act) {
+      const u16* kr = kbase + (size_t)p * hd + e0;
+      dot = q0 * b2f(kr[0]) + q1 * b2f(kr[1]);
+    }
+    dot = wave_sum(dot) * scale;
+    dot = __shfl(dot, 0, WAVE);
+    float mn = fmaxf(m, dot);
+    float alpha = __expf(m - mn);
+    float pw = __expf(dot - mn);
+    float v0 = 0.f, v1 = 0.f;
+    if (act) {
+      const u16* vr = vbase + (size_t)p * hd + e0;
+      v0 = b2f(vr[0]);
+      v1 = b2f(vr[1]);
+    }
+    o0 = o0 * alpha + pw * v0;
+    o1 = o1 * alpha + pw * v1;
+    l = l * alpha + pw;
+    m = mn;
+  }
+  if (act) {
+    u16* orow = out + (size_t)sidx * out_stride + (size_t)h * hd + e0;
+    orow[0] = f2b(o0 / l);
+    orow[1] = f2b(o1 / l);
+  }
+}
+
+// standalone rope for the op-level parity surface: x (BH, S, D) bf16 in
+// place, cos/sin (S, D/2) f32 (backends/mod.rs:444-482 layout)
+__global__ void k_rope_simple(u16* __restrict__ x,
+                              const float* __restrict__ cost,
+                              const float* __restrict__ sint, int s, int d) {
+  const int row = blockIdx.x;  // (bh, s) flattened
+  const int si = row % s;
+  const int half = d / 2;
+  u16* xr = x + (size_t)row * d;
+  const float* c = cost + (size_t)si * half;
+  const float* sn = sint + (size_t)si * half;
+  for (int i = threadIdx.x; i < half; i += blockDim.x) {
+    float x1 = b2f(xr[i]), x2 = b2f(xr[i + half]);
+    xr[i] = f2b(x1 * c[i] - x2 * sn[i]);
+    xr[i + half] = f2b(x2 * c[i] + x1 * sn[i]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// argmax (greedy sampling, text_model.rs:104): first-index tie-break to
+// match np.argmax / candle ArgMax.  Two passes over f32 logits.
+// ---------------------------------------------------------------------------
+__global__ void k_argmax_part(const float* __restrict__ logits, int n,
+                              float* __restrict__ pval, int* __restrict__ pidx,
+                              int nparts) {
+  const int part = blockIdx.x;
+  const int span = (n + nparts - 1) / nparts;
+  const int start = part * span, end = min(start + span, n);
+  float best = -INFINITY;
+  int bidx = 0x7fffffff;
+  for (int i = start + (int)threadIdx.x; i < end; i += blockDim.x) {
+    float v = logits[i];
+    if (v > best || (v == best && i < bidx)) { best = v; bidx = i; }
+  }
+  // wave+block reduce keeping first index on ties
+  __shared__ float sv[256];
+  __shared__ int si[256];
+  sv[threadIdx.x] = best;
+  si[threadIdx.x] = bidx;
+  __syncthreads();
+  for (int off = blockDim.x / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      float ov = sv[threadIdx.x + off];
+      int oi = si[threadIdx.x + off];
+      if (ov > sv[threadIdx.x] ||
+          (ov == sv[threadIdx.x] && oi < si[threadIdx.x])) {
+        sv[threadIdx.x] = ov;
+        si[threadIdx.x] = oi;
+      }
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) { pval[part] = sv[0]; pidx[part] = si[0]; }
+}
+
+// Final reduce; also appends the winning token to the output ring and
+// advances the device position (end-of-decode-step bookkeeping).
+__global__ void k_argmax_fin(const float* __restrict__ pval,
+                             const int* __restrict__ pidx, int nparts,
+                             u32* __restrict__ tok, int* __restrict__ pos,
+                             u32* __restrict__ ring, int* __restrict__ step,
+                             int advance_pos) {
+  __shared__ float sv[256];
+  __shared__ int si[256];
+  float best = -INFINITY;
+  int bidx = 0x7fffffff;
+  for (int i = threadIdx.x; i < nparts; i += blockDim.x) {
+    float v = pval[i];
+    if (v > best || (v == best && pidx[i] < bidx)) { best = v; bidx = pidx[i]; }
+  }
+  sv[threadIdx.x] = best;
+  si[threadIdx.x] = bidx;
+  __syncthreads();
+  for (int off = blockDim.x / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      float ov = sv[threadIdx.x + off];
+      int oi = si[threadIdx.x + off];
+      if (ov > sv[threadIdx.x] ||
+          (ov == sv[threadIdx.x] && oi < si[threadIdx.x])) {
+        sv[threadIdx.x] = ov;
+        si[threadIdx.x] = oi;
+      }
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    *tok = (u32)si[0];
+    if (ring) {
+      ring[*step] = (u32)si[0];
+      *step += 1;
+    }
+    if (advance_pos) *pos += 1;
+  }
+}
+
+__global__ void k_advance_pos(int* __restrict__ pos, int by) {
+  if (threadIdx.x == 0) *pos += by;
+}
+
+// ---------------------------------------------------------------------------
+// Prefill GEMM — MFMA bf16.  C[M,N] = A[M,K] @ W[N,K]^T (+ residual).
+// 128x128 tile, BK=64, 256 threads (4 waves as 2x2 of 64x64 sub-tiles),
+// v_mfma_f32_16x16x32_bf16, double-buffered LDS filled by
+// global_load_lds_dwordx4 with the XOR source-swizzle (rule 21) so the
+// fragment ds_read_b128 is bank-conflict-free (guide §5 ladder step 3).
+// Row clamp handles M/N tails; requires K % 64 == 0 (all hot-path K are).
+// ---------------------------------------------------------------------------
+#define GEMM_BM 128
+#define GEMM_BN 128
+#define GEMM_BK 64
+
+template <int EPI>  // 0: bf16 store, 1: bf16 store + residual add
+__global__ __launch_bounds__(256) void k_gemm_bf16(
+    const u16* __restrict__ A, const u16* __restrict__ W, u16* __restrict__ C,
+    const u16* __restrict__ res, int M, int N, int K) {
+  // dynamic-LDS base must be 16-B aligned (guide G17)
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  u16* As = reinterpret_cast<u16*>(smem);                       // [2][128][64]
+  u16* Bs = reinterpret_cast<u16*>(smem + 2 * GEMM_BM * GEMM_BK * 2);
+
+  const int m0 = blockIdx.x * GEMM_BM;
+  const int n0 = blockIdx.y * GEMM_BN;
+  const int t = threadIdx.x;
+  const int wid = t / WAVE, lane = t % WAVE;
+  const int wr = wid / 2, wc = wid % 2;  // wave's 64x64 quadrant
+
+  // staging geometry: each wave issues 4 glds per 16 KB tile; instruction j
+  // of wave w writes LDS rows [(w*4+j)*8, +8) (8 rows x 128 B), lane l ->
+  // row sub = l/8, 16-B unit u = l%8, with source column unit u ^ (row & 7).
+  auto stage = [&](int buf, int kt) {
+    const int ktbase = kt * GEMM_BK;
+    u16* as = As + (size_t)buf * GEMM_BM * GEMM_BK;
+    u16* bs = Bs + (size_t)buf * GEMM_BN * GEMM_BK;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int r = (wid * 4 + j) * 8 + lane / 8;
+      const int u = (lane % 8) ^ (r & 7);
+      {  // A tile
+        int grow = min(m0 + r, M - 1);
+        const u16* src = A + (size_t)grow * K + ktbase + u * 8;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)(uintptr_t)src,
+            (__attribute__((address_space(3))) void*)(uintptr_t)(
+                as + (size_t)(wid * 4 + j) * 8 * GEMM_BK),
+            16, 0, 0);
+      }
+      {  // W tile (B^T)
+        int grow = min(n0 + r, N - 1);
+        const u16* src = W + (size_t)grow * K + ktbase + u * 8;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)(uintptr_t)src,
+            (__attribute__((address_space(3))) void*)(uintptr_t)(
+                bs + (size_t)(wid * 4 + j) * 8 * GEMM_BK),
+            16, 0, 0);
+      }
+    }
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ntiles = K / GEMM_BK;
+  stage(0, 0);
+  __syncthreads();  // drains the glds (vmcnt(0) implied by the barrier)
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < ntiles) stage(cur ^ 1, kt + 1);
+    const u16* as = As + (size_t)cur * GEMM_BM * GEMM_BK;
+    const u16* bs = Bs + (size_t)cur * GEMM_BN * GEMM_BK;
+#pragma unroll
+    for (int ks = 0; ks < GEMM_BK / 32; ++ks) {
+      bf16x8 af[4], bf[4];
+      const int kk = ks * 32 + (lane / 16) * 8;
+      const int ku = kk / 8;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int ar = wr * 64 + i * 16 + (lane % 16);
+        af[i] = *reinterpret_cast<const bf16x8*>(
+            as + (size_t)ar * GEMM_BK + (size_t)(ku ^ (ar & 7)) * 8);
+        const int br = wc * 64 + i * 16 + (lane % 16);
+        bf[i] = *reinterpret_cast<const bf16x8*>(
+            bs + (size_t)br * GEMM_BK + (size_t)(ku ^ (br & 7)) * 8);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C/D map (guide §3): col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wr * 64 + i * 16 + (lane / 16) * 4 + r;
+        const int col = n0 + wc * 64 + j * 16 + (lane % 16);
+        if (row < M && col < N) {
+          float v = acc[i][j][r];
+          if (EPI == 1) v += b2f(res[(size_t)row * N + col]);
+          C[(size_t)row * N + col] = f2b(v);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// launch wrappers (host side) — declared in kernels.h, used by engine.hip
+// ---------------------------------------------------------------------------
+#include "kernels.h"
+
+void launch_f32_to_bf16(const float* in, u16* out, size_t n, hipStream_t s) {
+  int blocks = (int)min((n + 255) / 256, (size_t)2048);
+  hipLaunchKernelGGL(k_f32_to_bf16, dim3(blocks), dim3(256), 0, s, in, out, n);
+}
+void launch_bf16_to_f32(const u16* in, float* out, size_t n, hipStream_t s) {
+  int blocks = (int)min((n + 255) / 256, (size_t)2048);
+  hipLaunchKernelGGL(k_bf16_to_f32, dim3(blocks), dim3(256), 0, s, in, out, n);
+}
+void launch_fill_random(u16* out, size_t n, uint64_t seed, float scale,
+                        hipStream_t s) {
+  hipLaunchKernelGGL(k_fill_random, dim3(2048), dim3(256), 0, s, out, n, seed,
+                     scale);
+}
+void launch_fill_const(u16* out, size_t n, float v, hipStream_t s) {
+  hipLaunchKernelGGL(k_fill_const, dim3(512), dim3(256), 0, s, out, n, v);
+}
+void launch_rmsnorm(const u16* x, const u16* w, u16* out, int rows, int cols,
+                    float eps, hipStream_t s) {
+  hipLaunchKernelGGL(k_rmsnorm, dim3(rows), dim3(256), 0, s, x, w, out, cols,
+                     eps, rows, (size_t)0);
+}
+void launch_rmsnorm_strided(const u16* x, const u16* w, u16* out, int outer,
+                            int inner, size_t outer_stride, int cols,
+                            float eps, hipStream_t s) {
+  hipLaunchKernelGGL(k_rmsnorm, dim3(outer * inner), dim3(256), 0, s, x, w,
+                     out, cols, eps, inner, outer_stride);
+}
+void launch_silu_mul_rows(const u16* gu, u16* out, int S, int I,
+                          hipStream_t s) {
+  size_t n = (size_t)S * I;
+  int blocks = (int)min((n + 255) / 256, (size_t)4096);
+  hipLaunchKernelGGL(k_silu_mul_rows, dim3(blocks), dim3(256), 0, s, gu, out,
+                     S, I);
+}
+void launch_silu_mul(const u16* g, const u16* u, u16* out, size_t n,
+                     hipStream_t s) {
+  int blocks = (int)min((n + 255) / 256, (size_t)4096);
+  hipLaunchKernelGGL(k_silu_mul, dim3(blocks), dim3(256), 0, s, g, u, out, n);
+}
+void launch_gemv(const u16* W, const u16* x, void* out, const u16* res, int N,
+                 int K, int epi, hipStream_t s) {
+  dim3 grid((N + 7) / 8);
+  if (epi == 0)
+    hipLaunchKernelGGL(k_gemv<0>, grid, dim3(256), 0, s, W, x, out, res, N, K);
+  else if (epi == 1)
+    hipLaunchKernelGGL(k_gemv<1>, grid, dim3(256), 0, s, W, x, out, res, N, K);
+  else
+    hipLaunchKernelGGL(k_gemv<2>, grid, dim3(256), 0, s, W, x, out, res, N, K);
+}
+void launch_gemv_gateup(const u16* W, const u16* x, u16* out, int I, int K,
+                        hipStream_t s) {
+  dim3 grid((I + 7) / 8);
+  hipLaunchKernelGGL(k_gemv_gateup, grid, dim3(256), 0, s, W, x, out, I, K);
+}
+void launch_embed_token(const u16* embed, const u32* tok, u16* x, int H,
+                        hipStream_t s) {
+  hipLaunchKernelGGL(k_embed_token, dim3(1), dim3(256), 0, s, embed, tok, x, H);
+}
+void launch_embed_rows(const u16* embed, const u32* ids, u16* x, int S, int H,
+                       hipStream_t s) {
+  hipLaunchKernelGGL(k_embed_rows, dim3(S), dim3(256), 0, s, embed, ids, x, H);
+}
+void launch_rope_store_decode(u16* qkv, u16* kc, u16* vc, const float* cost,
+                              const float* sint, const int* pos, int nh,
+                              int nkv, int hd, int rd, int max_seq,
+                              hipStream_t s) {
+  hipLaunchKernelGGL(k_rope_store_decode, dim3(nh + 2 * nkv), dim3(64), 0, s,
+                     qkv, kc, vc, cost, sint, pos, nh, nkv, hd, rd, max_seq);
+}
+void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, const float* cost,
+                               const float* sint, int pos0, int S, int nh,
+                               int nkv, int hd, int rd, int max_seq,
+                               int qkv_stride, hipStream_t s) {
+  hipLaunchKernelGGL(k_rope_store_prefill, dim3(nh + 2 * nkv, S), dim3(64), 0,
+                     s, qkv, kc, vc, cost, sint, pos0, nh, nkv, hd, rd,
+                     max_seq, qkv_stride);
+}
+void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
+                        const int* pos, float* ws, u16* out, int nh, int nkv,
+                        int hd, int max_seq, int nchunk, hipStream_t s) {
+  hipLaunchKernelGGL(k_attn_decode_part, dim3(nchunk, nh), dim3(256), 0, s, q,
+                     kc, vc, pos, ws, nh, nkv, hd, max_seq, nchunk);
+  hipLaunchKernelGGL(k_attn_decode_combine, dim3(nh), dim3(128), 0, s, ws, out,
+                     nh, hd, nchunk);
+}
+void launch_attn_prefill(const u16* qkv, const u16* kc, const u16* vc,
+                         u16* out, int S, int pos0, int nh, int nkv, int hd,
+                         int max_seq, int qkv_stride, int out_stride,
+                         hipStream_t s) {
+  hipLaunchKernelGGL(k_attn_prefill, dim3((S + 3) / 4, nh), dim3(256), 0, s,
+                     qkv, kc, vc, out, S, pos0, nh, nkv, hd, max_seq,
+                     qkv_stride, out_stride);
+}
+void launch_rope_simple(u16* x, const float* cost, const float* sint, int bh,
+                        int s, int d, hipStream_t st) {
+  hipLaunchKernelGGL(k_rope_simple, dim3(bh * s), dim3(64), 0, st, x, cost,
+                     sint, s, d);
+}
+void launch_argmax(const float* logits, int n, float* pval, int* pidx,
+                   u32* tok, int* pos, u32* ring, int* step, int advance_pos,
+                   hipStream_t s) {
+  const int nparts = 256;
+  hipLaunchKernelGGL(k_argmax_part, dim3(nparts), dim3(256), 0, s, logits, n,
+                     pval, pidx, nparts);
+  hipLaunchKernelGGL(k_argmax_fin, dim3(1), dim3(256), 0, s, pval, pidx,
+                     nparts, tok, pos, ring, step, advance_pos);
+}
+void launch_advance_pos(int* pos, int by, hipStream_t s) {
+  hipLaunchKernelGGL(k_advance_pos, dim3(1), dim3(64), 0, s, pos, by);
+}
+void launch_gemm(const u16* A, const u16* W, u16* C, const u16* res, int M,
+                 int N, int K, int epi, hipStream_t s) {
+  dim3 grid((M + GEMM_BM - 1) / GEMM_BM, (N + GEMM_BN - 1) / GEMM_BN);
+  size_t lds = 2 * (size_t)(GEMM_BM + GEMM_BN) * GEMM_BK * 2;
+  if (epi == 0)
+    hipLaunchKernelGGL(k_gemm_bf16<0>, grid, dim3(256), lds, s, A, W, C, res,
+                       M, N, K);
+  else
+    hipLaunchKernelGGL(k_gemm_bf16<1>, grid, dim3(256), lds, s, A, W, C, res,
+                       M, N, K);
+}

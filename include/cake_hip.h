@@ -1,0 +1,159 @@
+/* cake_hip — C-ABI drop-in boundary for cake's layer-sharded LLM hot path,
+ * implemented MI355X-native (hand-written HIP/CDNA4 kernels for gfx950,
+ * RCCL point-to-point over xGMI for the inter-shard activation hop).
+ *
+ * This header is the FFI seam a Rust host (cake-core) would dlopen/bind,
+ * mirroring the seam cake already proved in backends/rocm/ffi.rs:12-51.
+ * Each entry point cites the reference interface it replaces
+ * (file:line into /root/reference/cake-core/src).  See INTEGRATION.md for
+ * the reference-side binding a cake maintainer would add.
+ *
+ * Conventions (mirrors backends/mod.rs:40-685 + cake/mod.rs:511-556):
+ *   - every function returns 0 on success, nonzero error code otherwise;
+ *     cake_hip_last_error() returns a human-readable message for the last
+ *     failure on this thread (mirrors the Result/anyhow error strings the
+ *     Forwarder contract uses, cake/mod.rs:520-533);
+ *   - plain pointers + sizes only, no torch/candle types;
+ *   - host-side f32 buffers at the boundary; the engine stores/computes
+ *     bf16 with f32 accumulation on device (matching cake's CUDA backend
+ *     dtype policy, attention.rs:270-277 + ops.cu f32 accumulators);
+ *   - all engine calls are synchronous at the API boundary unless noted
+ *     (ComputeBackend::synchronize semantics, backends/mod.rs:682-684).
+ */
+#ifndef CAKE_HIP_H
+#define CAKE_HIP_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef struct cake_engine cake_engine;
+
+/* ---- error reporting (mirrors backends/rocm/ffi.rs error-seam) ---------- */
+const char *cake_hip_last_error(void);
+
+/* ---- engine lifecycle ----------------------------------------------------
+ * Replaces Context::from_args + TextModelBase::load
+ * (cake/mod.rs:114-507, models/common/text_model.rs:150-263).
+ *
+ * config_json: contents of an HF-style config.json (the same file cake
+ *   auto-detects, cake/mod.rs:82-110,268-274).
+ * layer_lo/layer_hi: this shard's contiguous transformer-layer range
+ *   [lo, hi) — cake's contiguous-range assignment (sharding/default.rs:11-130).
+ * flags: CAKE_HIP_HAS_EMBED | CAKE_HIP_HAS_HEAD on the master-analog rank
+ *   (embedding + lm_head live on rank 0, text_model.rs:158-193).
+ * max_seq: KV-cache capacity in tokens (replaces cake's cat-per-token
+ *   cache, cache.rs:195-196, with a preallocated device cache).
+ * max_batch_tokens: largest prefill chunk (activation workspace rows).
+ * device: HIP device ordinal for this process.
+ */
+enum {
+  CAKE_HIP_HAS_EMBED = 1,
+  CAKE_HIP_HAS_HEAD = 2,
+  CAKE_HIP_USE_GRAPH = 4,   /* capture the decode step as a hipGraph */
+  CAKE_HIP_STATS = 8,       /* per-kernel hipEvent timing (bench roofline) */
+};
+
+int cake_hip_engine_create(const char *config_json, int layer_lo,
+                           int layer_hi, int flags, int max_seq,
+                           int max_batch_tokens, int device,
+                           cake_engine **out);
+
+/* Resolve a cake topology YAML (sharding/topology.rs:134-169, including
+ * "model.layers.0-15" range expressions, topology.rs:13,142-166) to the
+ * contiguous layer range of `node_name`.  Returns an error if the node's
+ * layers are non-contiguous (the build keeps cake's contiguous-range
+ * model, SURVEY.md §2 topology row). */
+int cake_hip_topology_node_range(const char *topology_yaml,
+                                 const char *node_name, int *lo, int *hi);
+
+void cake_hip_engine_free(cake_engine *e);
+
+/* ---- weight loading ------------------------------------------------------
+ * Replaces the safetensors VarBuilder path (utils/mod.rs:251-370): mmap the
+ * file, select this shard's tensors by the "model.layers.N." prefix, and
+ * upload straight to HBM as bf16.  F32 and BF16 source dtypes accepted. */
+int cake_hip_load_safetensors(cake_engine *e, const char *path);
+
+/* Seeded random init on device (synthetic-weights benches; no network for
+ * checkpoints).  RMS weights = 1, linear weights ~ scale * uniform(-1,1). */
+int cake_hip_init_random(cake_engine *e, uint64_t seed, float scale);
+
+/* ---- generation ----------------------------------------------------------
+ * Replaces TextModelBase::next_token / forward
+ * (text_model.rs:266-368,397-495): greedy = ArgMax (text_model.rs:104).
+ *
+ * prefill: run the whole prompt (index_pos 0 after reset, or appended at the
+ * current position), return the next greedy token and optionally the full
+ * last-position logits (f32, vocab_size).  Only meaningful on the rank with
+ * HAS_EMBED|HAS_HEAD; other ranks pass tokens=NULL and participate via the
+ * RCCL pipeline.  n_tokens must match across ranks. */
+int cake_hip_prefill(cake_engine *e, const uint32_t *tokens, int n_tokens,
+                     uint32_t *next_token, float *logits_out);
+
+/* decode: run `steps` greedy KV-cached decode steps from the current state
+ * (context size 1 per step, text_model.rs:400-420).  tokens_out (rank 0
+ * only, may be NULL on other ranks) receives the generated ids. */
+int cake_hip_decode(cake_engine *e, int steps, uint32_t *tokens_out);
+
+/* Clear KV cache + position — Goodbye semantics (worker.rs:364-384,
+ * cache.rs:248-253). */
+int cake_hip_reset(cake_engine *e);
+
+/* ---- Forwarder mirror ----------------------------------------------------
+ * Forwarder::forward / forward_batch (cake/mod.rs:519-546): run this
+ * shard's blocks [layer_lo, layer_hi) on host-f32 hidden states
+ * (seq, hidden), with KV appended at index_pos.  This is the unit the
+ * worker serve-loop executes per Message::Batch (sharding/worker.rs:299-578)
+ * and what the parity tests drive. */
+int cake_hip_forward_hidden(cake_engine *e, const float *x, int seq,
+                            int index_pos, float *out);
+
+/* ---- cluster transport ---------------------------------------------------
+ * Replaces the TCP/zstd wire hop (sharding/client.rs:79-174 +
+ * proto/message.rs) with an RCCL communicator over xGMI: the per-token
+ * activation exchange becomes ncclSend/ncclRecv between adjacent ranks. */
+#define CAKE_HIP_COMM_ID_BYTES 128
+int cake_hip_comm_id(uint8_t out[CAKE_HIP_COMM_ID_BYTES]);
+int cake_hip_comm_init(cake_engine *e, int rank, int world_size,
+                       const uint8_t id[CAKE_HIP_COMM_ID_BYTES]);
+
+/* ---- op-level surface (ComputeBackend mirror, for kernel parity tests) ---
+ * Host f32 in/out; the engine quantizes to bf16, runs the gfx950 kernel,
+ * and returns f32 — tests compare against oracle/ on bf16-quantized inputs.
+ * These are NOT the product generation path; they exercise the same kernels
+ * the path uses. */
+/* backends/mod.rs:244-246 */
+int cake_hip_op_rms_norm(int rows, int cols, float eps, const float *x,
+                         const float *w, float *out, int device);
+/* backends/mod.rs:206-241 — out[m,n] = sum_k x[m,k] * w[n,k]; w is (N,K) */
+int cake_hip_op_linear(int M, int N, int K, const float *x, const float *w,
+                       float *out, int device);
+/* backends/mod.rs:82 + ops.cu:101-138 */
+int cake_hip_op_silu_mul(long n, const float *gate, const float *up,
+                         float *out, int device);
+/* backends/mod.rs:444-482 — x (B,H,S,D), cos/sin (S, D/2) */
+int cake_hip_op_rope(int b, int h, int s, int d, const float *x,
+                     const float *cosv, const float *sinv, float *out,
+                     int device);
+
+/* ---- observability (bench roofline evidence) ----------------------------
+ * JSON {"kernels": {name: {"launches": n, "ms": t, "bytes": b}}} — per-
+ * kernel-family hipEvent timing + algorithmic bytes, collected while
+ * CAKE_HIP_STATS was set (SURVEY.md §5 metrics row). */
+int cake_hip_kernel_stats(cake_engine *e, char *buf, int cap);
+int cake_hip_stats_reset(cake_engine *e);
+int cake_hip_set_stats(cake_engine *e, int enabled);
+
+int cake_hip_sync(cake_engine *e);
+
+/* build info: returns "gfx950;<compile date>" */
+const char *cake_hip_build_info(void);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* CAKE_HIP_H */
