@@ -1,0 +1,298 @@
+#!/usr/bin/env python3
+"""Benchmark — cake's headline metric (decode tok/s + prefill tok/s) on the
+MI355X-native engine, per BASELINE.json.
+
+Contract: `python bench.py --gpus N --steps K --warmup W`; for N>1 launched
+by torchrun with one rank per GPU; rank 0 prints ONE JSON line.  A "step" is
+one greedy KV-cached decode token (the reference's per-token hot loop,
+text_model.rs:397-495; rate convention excludes the first token like
+master.rs:131-166).  N>1 = the model layer-sharded as contiguous pipeline
+ranges with the activation hop over RCCL/xGMI (cake's sharding model,
+default.rs:11-130) — total work fixed, so "scaling": "strong".
+
+Synthetic data: seeded-uniform token ids (seed 299792458 = cake's default,
+lib.rs:180); random-init weights of the named architecture (no network).
+The oracle (CPU restatement) is timed beside it as cpu_baseline — baseline
+only, never the target.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+SEED = 299792458
+
+# algorithmic bytes: streamed once per decode step
+HBM_PEAK_GBS = 8000.0  # MI355X spec peak (MI355X_MICROARCH.md)
+
+
+def log(rank, *a):
+    if rank == 0:
+        print(*a, file=sys.stderr, flush=True)
+
+
+def cpu_baseline(cfg_json, sample_layers=8, max_secs=45.0):
+    """Oracle (numpy restatement, kind='port') decode rate on this host.
+
+    Sample: greedy decode steps of the same architecture with an 8-token
+    context, weights stored f16 (converted f32 per op like the oracle's
+    astype), with `sample_layers` distinct random layers reused cyclically to
+    bound RAM (cache-neutral: one block is far larger than any L3).
+    Rate is scaled to a full-model token.
+    """
+    from oracle import Config, LayerWeights, ModelWeights, OracleModel
+    cfg = Config.from_json(cfg_json)
+    H, I, V = cfg.hidden_size, cfg.intermediate_size, cfg.vocab_size
+    hd, nh, nkv = cfg.hd, cfg.num_attention_heads, cfg.num_key_value_heads
+    rng = np.random.default_rng(SEED)
+
+    def t(*shape):
+        return (rng.standard_normal(shape, dtype=np.float32) * 0.02
+                ).astype(np.float16)
+
+    nlay = min(sample_layers, cfg.num_hidden_layers)
+    layers = [LayerWeights(
+        input_layernorm=np.ones(H, np.float16),
+        post_attention_layernorm=np.ones(H, np.float16),
+        q_proj=t(nh * hd, H), k_proj=t(nkv * hd, H), v_proj=t(nkv * hd, H),
+        o_proj=t(H, nh * hd), gate_proj=t(I, H), up_proj=t(I, H),
+        down_proj=t(H, I),
+        q_norm=np.ones(hd, np.float16) if cfg.use_qk_norm else None,
+        k_norm=np.ones(hd, np.float16) if cfg.use_qk_norm else None,
+    ) for _ in range(nlay)]
+    embed = t(V, H)
+    w = ModelWeights(embed_tokens=embed, norm=np.ones(H, np.float16),
+                     lm_head=embed if cfg.tie_word_embeddings else t(V, H),
+                     layers=[layers[i % nlay]
+                             for i in range(cfg.num_hidden_layers)])
+    # distinct KV per layer index is handled by OracleModel's kv list
+    model = OracleModel(cfg, w)
+    prompt = list(rng.integers(0, V, size=8))
+    model.generate_greedy(prompt, 1)  # includes prefill of the 8-token ctx
+    # timed steps
+    steps = 0
+    t0 = time.perf_counter()
+    while True:
+        logits = model.forward(np.array([[steps % V]], dtype=np.int64),
+                               9 + steps)
+        steps = steps + 1 if logits is not None else steps
+        del logits
+        el = time.perf_counter() - t0
+        if el > max_secs or steps >= 4:
+            break
+    try:
+        import multiprocessing
+        cores = multiprocessing.cpu_count()
+    except Exception:
+        cores = 1
+    return {
+        "value": steps / el, "unit": "tok/s", "cores": cores, "kind": "port",
+        "sample": (f"{steps} greedy decode steps, 8-token context, f16 "
+                   f"weights, {nlay} distinct layers reused cyclically"),
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=256)
+    ap.add_argument("--warmup", type=int, default=16)
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--prompt-len", type=int, default=128)
+    ap.add_argument("--prefill-len", type=int, default=2048,
+                    help="tokens for the prefill-rate measurement (0=skip)")
+    ap.add_argument("--max-seq", type=int, default=4096)
+    ap.add_argument("--no-graph", action="store_true")
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--stats-steps", type=int, default=8)
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(world, args.gpus)
+    if world > 1 and args.gpus != world:
+        args.gpus = world
+
+    import cake_amd
+    from cake_amd.configs import MODELS, weight_bytes_bf16
+    cfg_json = MODELS[args.model]
+    L = cfg_json["num_hidden_layers"]
+
+    dist = None
+    if world > 1:
+        import torch
+        import torch.distributed as tdist
+        dist = tdist
+        dist.init_process_group("gloo")
+
+    # contiguous pipeline ranges via the cake topology boundary
+    # (topology.rs range expressions; equal split like default.rs would give
+    # equal-TFLOPS workers)
+    bounds = [round(L * r / world) for r in range(world + 1)]
+    topo_yaml = "".join(
+        f"rank{r}:\n  host: 127.0.0.1:{10128 + r}\n  layers:\n"
+        f"    - model.layers.{bounds[r]}-{bounds[r + 1] - 1}\n"
+        for r in range(world))
+    flags = 0
+    if rank == 0:
+        flags |= cake_amd.HAS_EMBED | cake_amd.HAS_HEAD
+    if not args.no_graph:
+        flags |= cake_amd.USE_GRAPH
+
+    log(rank, f"[bench] creating engine model={args.model} world={world} "
+        f"rank={rank} layers={bounds[rank]}..{bounds[rank + 1]}")
+    eng = cake_amd.Engine.from_topology(
+        json.dumps(cfg_json), topo_yaml, f"rank{rank}", flags=flags,
+        max_seq=args.max_seq, max_batch_tokens=2048, device=local_rank)
+    t0 = time.perf_counter()
+    eng.init_random(seed=SEED, scale=0.02)
+    log(rank, f"[bench] random-init weights "
+        f"({weight_bytes_bf16(cfg_json, bounds[rank], bounds[rank + 1], rank == 0, rank == 0) / 1e9:.2f} GB shard) "
+        f"in {time.perf_counter() - t0:.2f}s")
+
+    if world > 1:
+        import torch
+        if rank == 0:
+            cid = cake_amd.comm_id()
+            tens = torch.tensor(bytearray(cid), dtype=torch.uint8)
+        else:
+            tens = torch.zeros(cake_amd.COMM_ID_BYTES, dtype=torch.uint8)
+        dist.broadcast(tens, src=0)
+        eng.comm_init(rank, world, bytes(tens.numpy().tobytes()))
+
+    rng = np.random.default_rng(SEED)
+    prompt = rng.integers(0, cfg_json["vocab_size"],
+                          size=args.prompt_len).astype(np.uint32)
+
+    def barrier_sync():
+        eng.sync()
+        if dist:
+            dist.barrier()
+
+    # ---- prefill the prompt + warmup ------------------------------------
+    if rank == 0:
+        eng.prefill(prompt)
+    else:
+        eng.prefill_participate(len(prompt))
+    if args.warmup > 0:
+        if rank == 0:
+            eng.decode(args.warmup)
+        else:
+            eng.decode_participate(args.warmup)
+    barrier_sync()
+
+    # ---- timed decode region (the headline metric) ----------------------
+    t0 = time.perf_counter()
+    if rank == 0:
+        toks = eng.decode(args.steps)
+    else:
+        eng.decode_participate(args.steps)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+    if dist:
+        import torch
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    decode_tok_s = args.steps / elapsed
+    log(rank, f"[bench] decode: {args.steps} steps in {elapsed:.3f}s = "
+        f"{decode_tok_s:.1f} tok/s")
+
+    # ---- roofline stats segment (eager, hipEvent-timed) ------------------
+    roofline = None
+    if args.stats_steps > 0:
+        eng.set_stats(True)
+        if rank == 0:
+            eng.decode(args.stats_steps)
+        else:
+            eng.decode_participate(args.stats_steps)
+        barrier_sync()
+        eng.set_stats(False)
+        if rank == 0:
+            st = eng.kernel_stats()["kernels"]
+            dom = max(st.items(), key=lambda kv: kv[1]["ms"])
+            name, d = dom
+            gbs = d["bytes"] / (d["ms"] * 1e-3) / 1e9 if d["ms"] > 0 else 0
+            roofline = {
+                "bound": "hbm", "achieved": round(gbs, 1),
+                "peak": HBM_PEAK_GBS, "unit": "GB/s",
+                "frac": round(gbs / HBM_PEAK_GBS, 4),
+                "traffic": None, "kernel": name,
+                "per_launch_ms": round(d["ms"] / d["launches"], 5),
+                "all_kernels": {k: {"ms": round(v["ms"], 3),
+                                    "launches": v["launches"],
+                                    "gbs": round(v["bytes"] / (v["ms"] * 1e-3)
+                                                 / 1e9, 1) if v["ms"] > 0
+                                    else 0}
+                                for k, v in st.items()},
+            }
+        eng.stats_reset()
+
+    # ---- prefill rate ----------------------------------------------------
+    prefill_tok_s = None
+    if args.prefill_len > 0 and args.prefill_len <= args.max_seq:
+        eng.reset()
+        pf = rng.integers(0, cfg_json["vocab_size"],
+                          size=args.prefill_len).astype(np.uint32)
+        barrier_sync()
+        t0 = time.perf_counter()
+        if rank == 0:
+            eng.prefill(pf)
+        else:
+            eng.prefill_participate(len(pf))
+        barrier_sync()
+        el = time.perf_counter() - t0
+        if dist:
+            import torch
+            t = torch.tensor([el], dtype=torch.float64)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            el = float(t.item())
+        prefill_tok_s = args.prefill_len / el
+        log(rank, f"[bench] prefill: {args.prefill_len} tokens in {el:.3f}s "
+            f"= {prefill_tok_s:.0f} tok/s")
+
+    # ---- CPU baseline (rank 0, N=1 only) ---------------------------------
+    cpu = None
+    if rank == 0 and world == 1 and not args.no_cpu_baseline:
+        log(rank, "[bench] timing CPU baseline (oracle, bounded sample) ...")
+        cpu = cpu_baseline(cfg_json)
+        log(rank, f"[bench] cpu_baseline: {cpu['value']:.3f} tok/s on "
+            f"{cpu['cores']} cores")
+
+    if rank == 0:
+        out = {
+            "metric": "decode tok/s",
+            "value": round(decode_tok_s, 2),
+            "unit": "tok/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 4),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "workload": f"{args.model}-bf16-decode",
+                "model": args.model,
+                "prompt_len": args.prompt_len,
+                "gen_tokens": args.steps,
+                "parallelism": f"pp{n_gpus}",
+            },
+            "prefill_tok_s": round(prefill_tok_s, 1) if prefill_tok_s else None,
+            "prefill_len": args.prefill_len,
+            "roofline": roofline,
+            "cpu_baseline": cpu,
+        }
+        print(json.dumps(out), flush=True)
+
+
+if __name__ == "__main__":
+    main()
