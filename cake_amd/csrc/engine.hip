@@ -223,12 +223,13 @@ struct cake_engine {
   float* pval = nullptr;
   int* pidx = nullptr;
   float* attn_ws = nullptr;
+  u32* attn_cnt = nullptr;
   int* dev_pos = nullptr;
   int* dev_step = nullptr;
   u32* dev_tok = nullptr;
   int host_pos = 0;
 
-  static const int NCHUNK = 16;
+  int nchunk = 8;  // split-KV chunks for decode attention (CAKE_NCHUNK)
 
   hipStream_t stream = nullptr;
   hipGraphExec_t graph = nullptr;
@@ -317,14 +318,11 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
   const ModelConfig& c = e->c;
   const int H = c.hidden, I = c.inter, hd = c.hd();
   const int Sq = c.sq(), Nq = c.nqkv();
-  {  // rms_1
-    StatScope ss(e, "rmsnorm", 2.0 * H * 2 + H * 2, 0);
-    launch_rmsnorm(e->x, l.rms1, e->xn, 1, H, c.rms_eps, e->stream);
-  }
-  {  // fused qkv projection (GEMV)
-    StatScope ss(e, "gemv_qkv", (double)Nq * H * 2 + H * 2 + Nq * 2,
+  {  // rms_1 fused into the qkv projection (GEMV, x kept in registers)
+    StatScope ss(e, "gemv_qkv", (double)Nq * H * 2 + 2.0 * H * 2 + Nq * 2,
                  2.0 * Nq * H);
-    launch_gemv(l.wqkv, e->xn, e->qkv, nullptr, Nq, H, 0, e->stream);
+    launch_gemv(l.wqkv, e->x, e->qkv, nullptr, l.rms1, c.rms_eps, Nq, H, 0,
+                e->stream);
   }
   if (l.qnorm) {  // Qwen3 per-head QK-norm (attention.rs:202-215)
     StatScope ss(e, "qknorm", 2.0 * (c.nh + c.nkv) * hd * 2, 0);
@@ -338,32 +336,31 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
                              e->dev_pos, c.nh, c.nkv, hd, hd, e->max_seq,
                              e->stream);
   }
-  {  // decode attention over the cache
+  {  // decode attention over the cache (single launch, split-KV combine)
     double kvbytes = 2.0 * (e->host_pos + 1) * c.skv() * 2;
     StatScope ss(e, "attn_decode", kvbytes + Sq * 2 * 2,
                  4.0 * (e->host_pos + 1) * Sq);
     launch_attn_decode(e->qkv, l.kc, l.vc, e->dev_pos, e->attn_ws,
-                       e->attn_out, c.nh, c.nkv, hd, e->max_seq, e->NCHUNK,
-                       e->stream);
+                       e->attn_cnt, e->attn_out, c.nh, c.nkv, hd, e->max_seq,
+                       e->nchunk, e->stream);
   }
   {  // o projection + residual
     StatScope ss(e, "gemv_o", (double)H * Sq * 2 + Sq * 2 + H * 4,
                  2.0 * H * Sq);
-    launch_gemv(l.wo, e->attn_out, e->x, e->x, H, Sq, 1, e->stream);
+    launch_gemv(l.wo, e->attn_out, e->x, e->x, nullptr, 0.f, H, Sq, 1,
+                e->stream);
   }
-  {  // rms_2
-    StatScope ss(e, "rmsnorm", 2.0 * H * 2 + H * 2, 0);
-    launch_rmsnorm(e->x, l.rms2, e->xn, 1, H, c.rms_eps, e->stream);
-  }
-  {  // fused gate_up GEMV + silu_mul (mlp.rs:21-31)
-    StatScope ss(e, "gemv_gateup", 2.0 * I * H * 2 + H * 2 + I * 2,
+  {  // rms_2 fused into gate_up GEMV + silu_mul (mlp.rs:21-31)
+    StatScope ss(e, "gemv_gateup", 2.0 * I * H * 2 + 2.0 * H * 2 + I * 2,
                  4.0 * I * H);
-    launch_gemv_gateup(l.wgu, e->xn, e->act, I, H, e->stream);
+    launch_gemv_gateup(l.wgu, e->x, e->act, l.rms2, c.rms_eps, I, H,
+                       e->stream);
   }
   {  // down projection + residual
     StatScope ss(e, "gemv_down", (double)H * I * 2 + I * 2 + H * 4,
                  2.0 * H * I);
-    launch_gemv(l.wdown, e->act, e->x, e->x, H, I, 1, e->stream);
+    launch_gemv(l.wdown, e->act, e->x, e->x, nullptr, 0.f, H, I, 1,
+                e->stream);
   }
 }
 
@@ -429,15 +426,12 @@ static void enqueue_layer_prefill(cake_engine* e, LayerDev& l, int S,
 static void enqueue_head_sample(cake_engine* e, int S, int advance_by) {
   const ModelConfig& c = e->c;
   const int H = c.hidden, V = c.vocab;
-  {  // final norm on the last token only (text_model.rs:336-346)
-    StatScope ss(e, "rmsnorm", 2.0 * H * 2, 0);
-    launch_rmsnorm(e->x + (size_t)(S - 1) * H, e->norm_w, e->xn, 1, H,
-                   c.rms_eps, e->stream);
-  }
-  {  // lm_head (f32 logits)
+  {  // final norm on the last token (text_model.rs:336-346) fused into
+     // the lm_head GEMV (f32 logits)
     StatScope ss(e, "gemv_head", (double)V * H * 2 + H * 2 + V * 4,
                  2.0 * V * H);
-    launch_gemv(e->lm_head, e->xn, e->logits, nullptr, V, H, 2, e->stream);
+    launch_gemv(e->lm_head, e->x + (size_t)(S - 1) * H, e->logits, nullptr,
+                e->norm_w, c.rms_eps, V, H, 2, e->stream);
   }
   if (advance_by > 1)
     launch_advance_pos(e->dev_pos, advance_by - 1, e->stream);
@@ -492,6 +486,8 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
   if (c.hd() > 128 || c.hd() % 8 != 0)
     return set_err(5, "head_dim %d unsupported (must be <=128, mult of 8)",
                    c.hd());
+  if (c.hidden > 16384)
+    return set_err(5, "hidden_size %d unsupported (> 16384)", c.hidden);
   if (max_seq <= 0) max_seq = c.max_pos;
   if (max_batch_tokens <= 0) max_batch_tokens = 2048;
   if (max_batch_tokens > max_seq) max_batch_tokens = max_seq;
@@ -558,7 +554,12 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
   ALLOC(e->ring, u32, cake_engine::RING_CAP);
   ALLOC(e->pval, float, 256);
   ALLOC(e->pidx, int, 256);
-  ALLOC(e->attn_ws, float, (size_t)c.nh * cake_engine::NCHUNK * (hd + 2));
+  if (const char* nc = getenv("CAKE_NCHUNK")) {
+    int v = atoi(nc);
+    if (v >= 1 && v <= 64) e->nchunk = v;
+  }
+  ALLOC(e->attn_ws, float, (size_t)c.nh * e->nchunk * (hd + 2));
+  ALLOC(e->attn_cnt, u32, c.nh);
   ALLOC(e->dev_pos, int, 1);
   ALLOC(e->dev_step, int, 1);
   ALLOC(e->dev_tok, u32, 1);
@@ -588,7 +589,8 @@ extern "C" void cake_hip_engine_free(cake_engine* e) {
   hipFree(e->x); hipFree(e->xn); hipFree(e->qkv); hipFree(e->attn_out);
   hipFree(e->gu); hipFree(e->act); hipFree(e->logits); hipFree(e->fbuf);
   hipFree(e->ids); hipFree(e->ring); hipFree(e->pval); hipFree(e->pidx);
-  hipFree(e->attn_ws); hipFree(e->dev_pos); hipFree(e->dev_step);
+  hipFree(e->attn_ws); hipFree(e->attn_cnt);
+  hipFree(e->dev_pos); hipFree(e->dev_step);
   hipFree(e->dev_tok); hipFree(e->cos_t); hipFree(e->sin_t);
   hipStreamDestroy(e->stream);
   delete e;
@@ -1097,7 +1099,7 @@ extern "C" int cake_hip_op_linear(int M, int N, int K, const float* x,
   if ((r = up16(x, o.fa, o.a, (size_t)M * K, o.s))) return r;
   if ((r = up16(w, o.fb, o.b, (size_t)N * K, o.s))) return r;
   if (M == 1) {
-    launch_gemv(o.b, o.a, o.c, nullptr, N, K, 0, o.s);
+    launch_gemv(o.b, o.a, o.c, nullptr, nullptr, 0.f, N, K, 0, o.s);
   } else {
     if (K % 64)
       return set_err(5, "gemm requires K %% 64 == 0 (got %d)", K);

@@ -21,10 +21,11 @@ void launch_silu_mul(const u16* g, const u16* u, u16* out, size_t n,
                      hipStream_t s);
 void launch_silu_mul_rows(const u16* gu, u16* out, int S, int I,
                           hipStream_t s);
-void launch_gemv(const u16* W, const u16* x, void* out, const u16* res, int N,
-                 int K, int epi, hipStream_t s);
-void launch_gemv_gateup(const u16* W, const u16* x, u16* out, int I, int K,
-                        hipStream_t s);
+void launch_gemv(const u16* W, const u16* x, void* out, const u16* res,
+                 const u16* nw, float eps, int N, int K, int epi,
+                 hipStream_t s);
+void launch_gemv_gateup(const u16* W, const u16* x, u16* out, const u16* nw,
+                        float eps, int I, int K, hipStream_t s);
 void launch_embed_token(const u16* embed, const u32* tok, u16* x, int H,
                         hipStream_t s);
 void launch_embed_rows(const u16* embed, const u32* ids, u16* x, int S, int H,
@@ -38,8 +39,9 @@ void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, const float* cost,
                                int nkv, int hd, int rd, int max_seq,
                                int qkv_stride, hipStream_t s);
 void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
-                        const int* pos, float* ws, u16* out, int nh, int nkv,
-                        int hd, int max_seq, int nchunk, hipStream_t s);
+                        const int* pos, float* ws, u32* cnt, u16* out, int nh,
+                        int nkv, int hd, int max_seq, int nchunk,
+                        hipStream_t s);
 void launch_attn_prefill(const u16* qkv, const u16* kc, const u16* vc,
                          u16* out, int S, int pos0, int nh, int nkv, int hd,
                          int max_seq, int qkv_stride, int out_stride,

@@ -49,6 +49,14 @@ __device__ __host__ inline float b2f(u16 h) {
   return v.f;
 }
 
+// nontemporal 16-B weight load: decode weights are streamed exactly once per
+// token — keep them from thrashing L2/L3 (guide G14: streaming data)
+__device__ inline short8 ntload8(const u16* p) {
+  uint4 v = __builtin_nontemporal_load(reinterpret_cast<const uint4*>(p));
+  union { uint4 u; short8 s; } c{v};
+  return c.s;
+}
+
 __device__ inline float wave_sum(float v) {
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, WAVE);
@@ -184,50 +192,94 @@ __global__ void k_silu_mul_rows(const u16* __restrict__ gu,
 }
 
 // ---------------------------------------------------------------------------
-// GEMV: y[r] = dot(W[r,:], x) for 8 rows per block.  Decode's workhorse —
-// HBM-bound weight streaming at 16 B/lane (guide §5 GEMV row: straight to
-// VGPRs, no LDS).  x is bf16 and L1/L2-resident (re-read per row).
-// EPI: 0 = bf16 out; 1 = bf16 out + residual add (res may alias out);
-//      2 = f32 out (lm_head logits)
+// GEMV family — decode's workhorse (HBM-bound weight streaming at 16 B/lane,
+// guide §5 "GEMV / M <= 16": weights straight to VGPRs, no LDS round trip).
+//
+// k_gemv_reg<ROWS, EPI, NORM, KB>: x is loaded ONCE into registers (up to
+// KB*8 f32/thread, K <= KB*2048) and reused for every row; with NORM the
+// kernel fuses the preceding rms_norm (backends/mod.rs:244-246) into the
+// x load — the normed value is re-quantized to bf16 so the fused path is
+// bit-identical to rmsnorm-then-gemv.  ROWS per block is chosen by the
+// launcher so the grid has >= ~2048 workgroups (256 CUs want many blocks).
+// EPI: 0 = bf16 out; 1 = bf16 out + residual add; 2 = f32 out (logits).
 // ---------------------------------------------------------------------------
-template <int EPI>
-__global__ __launch_bounds__(256) void k_gemv(
-    const u16* __restrict__ W, const u16* __restrict__ x, void* __restrict__ out,
-    const u16* __restrict__ res, int N, int K) {
+template <int ROWS, int EPI, bool NORM, int KB>
+__global__ __launch_bounds__(256) void k_gemv_reg(
+    const u16* __restrict__ W, const u16* __restrict__ x,
+    void* __restrict__ out, const u16* __restrict__ res,
+    const u16* __restrict__ nw, float eps, int N, int K) {
   const int t = threadIdx.x;
-  const int row0 = blockIdx.x * 8;
-  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-  const int kiter = (K + 2047) / 2048;  // 256 threads * 8 bf16
+  const int row0 = blockIdx.x * ROWS;
+  const int wid = t / WAVE, lane = t % WAVE;
+  __shared__ float red[ROWS > 4 ? ROWS : 4][4];
+
+  // phase 1: x -> registers (f32), optionally fused rms_norm
+  float xr[KB * 8];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    if (k0 < K) {
+      short8 xv = *reinterpret_cast<const short8*>(x + k0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xv[j]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = 0.f;
+    }
+  }
+  if (NORM) {
+    float ss = 0.f;
+#pragma unroll
+    for (int i = 0; i < KB * 8; ++i) ss += xr[i] * xr[i];
+    ss = wave_sum(ss);
+    if (lane == 0) red[0][wid] = ss;
+    __syncthreads();
+    const float scale =
+        rsqrtf((red[0][0] + red[0][1] + red[0][2] + red[0][3]) / (float)K +
+               eps);
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) {
+        short8 wv = *reinterpret_cast<const short8*>(nw + k0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          // re-quantize so fused == rmsnorm-kernel-then-gemv bit-exactly
+          xr[i * 8 + j] = b2f(f2b(xr[i * 8 + j] * scale * b2f((u16)wv[j])));
+      }
+    }
+  }
+
+  // phase 2: stream ROWS weight rows
+  float acc[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) acc[r] = 0.f;
 #pragma unroll 1
-  for (int r = 0; r < 8; ++r) {
+  for (int r = 0; r < ROWS; ++r) {
     const int row = row0 + r;
     if (row >= N) break;
     const u16* wr = W + (size_t)row * K;
     float a = 0.f;
-    for (int i = 0; i < kiter; ++i) {
-      int k0 = i * 2048 + t * 8;
-      if (k0 + 8 <= K) {
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) {
         short8 wv = *reinterpret_cast<const short8*>(wr + k0);
-        short8 xv = *reinterpret_cast<const short8*>(x + k0);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          a = fmaf(b2f((u16)wv[j]), b2f((u16)xv[j]), a);
-      } else {
-        for (int k = k0; k < K; ++k) a = fmaf(b2f(wr[k]), b2f(x[k]), a);
+          a = fmaf(b2f((u16)wv[j]), xr[i * 8 + j], a);
       }
     }
     acc[r] = a;
   }
-  // cross-wave reduce: 4 waves x 8 rows
-  __shared__ float red[8][4];
-  const int wid = t / WAVE, lane = t % WAVE;
 #pragma unroll
-  for (int r = 0; r < 8; ++r) {
+  for (int r = 0; r < ROWS; ++r) {
     float v = wave_sum(acc[r]);
     if (lane == 0) red[r][wid] = v;
   }
   __syncthreads();
-  if (t < 8) {
+  if (t < ROWS) {
     const int row = row0 + t;
     if (row < N) {
       float v = red[t][0] + red[t][1] + red[t][2] + red[t][3];
@@ -242,16 +294,113 @@ __global__ __launch_bounds__(256) void k_gemv(
   }
 }
 
-// fused gate_up GEMV + silu_mul: block computes out[i] = silu(g_i) * u_i for
-// 8 output channels, where g_i = dot(W[i,:], x), u_i = dot(W[i+I,:], x)
-// (mlp.rs:21-31 with the fused gate_up weight of mlp.rs:38-46)
+// streaming fallback for K > 16384 (e.g. 70B down-proj K=28672): x re-read
+// from L1/L2 per row
+template <int ROWS, int EPI>
+__global__ __launch_bounds__(256) void k_gemv_stream(
+    const u16* __restrict__ W, const u16* __restrict__ x,
+    void* __restrict__ out, const u16* __restrict__ res, int N, int K) {
+  const int t = threadIdx.x;
+  const int row0 = blockIdx.x * ROWS;
+  const int wid = t / WAVE, lane = t % WAVE;
+  float acc[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) acc[r] = 0.f;
+  const int kiter = (K + 2047) / 2048;
+#pragma unroll 1
+  for (int r = 0; r < ROWS; ++r) {
+    const int row = row0 + r;
+    if (row >= N) break;
+    const u16* wr = W + (size_t)row * K;
+    float a = 0.f;
+    for (int i = 0; i < kiter; ++i) {
+      int k0 = i * 2048 + t * 8;
+      if (k0 + 8 <= K) {
+        short8 wv = ntload8(wr + k0);
+        short8 xv = *reinterpret_cast<const short8*>(x + k0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          a = fmaf(b2f((u16)wv[j]), b2f((u16)xv[j]), a);
+      } else {
+        for (int k = k0; k < K; ++k) a = fmaf(b2f(wr[k]), b2f(x[k]), a);
+      }
+    }
+    acc[r] = a;
+  }
+  __shared__ float red[ROWS > 4 ? ROWS : 4][4];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) {
+    float v = wave_sum(acc[r]);
+    if (lane == 0) red[r][wid] = v;
+  }
+  __syncthreads();
+  if (t < ROWS) {
+    const int row = row0 + t;
+    if (row < N) {
+      float v = red[t][0] + red[t][1] + red[t][2] + red[t][3];
+      if (EPI == 2) {
+        reinterpret_cast<float*>(out)[row] = v;
+      } else if (EPI == 1) {
+        reinterpret_cast<u16*>(out)[row] = f2b(v + b2f(res[row]));
+      } else {
+        reinterpret_cast<u16*>(out)[row] = f2b(v);
+      }
+    }
+  }
+}
+
+// fused [rms_norm ->] gate_up GEMV -> silu_mul: block computes
+// out[i] = silu(g_i) * u_i for 8 channels, g_i = dot(W[i,:], xn),
+// u_i = dot(W[i+I,:], xn)  (mlp.rs:21-31 + fused gate_up of mlp.rs:38-46)
+template <bool NORM, int KB>
 __global__ __launch_bounds__(256) void k_gemv_gateup(
-    const u16* __restrict__ W, const u16* __restrict__ x, u16* __restrict__ out,
-    int I, int K) {
+    const u16* __restrict__ W, const u16* __restrict__ x,
+    u16* __restrict__ out, const u16* __restrict__ nw, float eps, int I,
+    int K) {
   const int t = threadIdx.x;
   const int c0 = blockIdx.x * 8;
-  float accg[8] = {0}, accu[8] = {0};
-  const int kiter = (K + 2047) / 2048;
+  const int wid = t / WAVE, lane = t % WAVE;
+  __shared__ float redg[8][4], redu[8][4];
+
+  float xr[KB * 8];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    if (k0 < K) {
+      short8 xv = *reinterpret_cast<const short8*>(x + k0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xv[j]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = 0.f;
+    }
+  }
+  if (NORM) {
+    float ss = 0.f;
+#pragma unroll
+    for (int i = 0; i < KB * 8; ++i) ss += xr[i] * xr[i];
+    ss = wave_sum(ss);
+    if (lane == 0) redg[0][wid] = ss;
+    __syncthreads();
+    const float scale =
+        rsqrtf((redg[0][0] + redg[0][1] + redg[0][2] + redg[0][3]) /
+                   (float)K + eps);
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) {
+        short8 wv = *reinterpret_cast<const short8*>(nw + k0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          xr[i * 8 + j] = b2f(f2b(xr[i * 8 + j] * scale * b2f((u16)wv[j])));
+      }
+    }
+  }
+
+  float accg[8], accu[8];
+#pragma unroll
+  for (int r = 0; r < 8; ++r) accg[r] = accu[r] = 0.f;
 #pragma unroll 1
   for (int r = 0; r < 8; ++r) {
     const int c = c0 + r;
@@ -259,30 +408,22 @@ __global__ __launch_bounds__(256) void k_gemv_gateup(
     const u16* wg = W + (size_t)c * K;
     const u16* wu = W + (size_t)(c + I) * K;
     float g = 0.f, u = 0.f;
-    for (int i = 0; i < kiter; ++i) {
-      int k0 = i * 2048 + t * 8;
-      if (k0 + 8 <= K) {
-        short8 xv = *reinterpret_cast<const short8*>(x + k0);
-        short8 gv = *reinterpret_cast<const short8*>(wg + k0);
-        short8 uv = *reinterpret_cast<const short8*>(wu + k0);
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) {
+        short8 gv = ntload8(wg + k0);
+        short8 uv = ntload8(wu + k0);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          float xf = b2f((u16)xv[j]);
-          g = fmaf(b2f((u16)gv[j]), xf, g);
-          u = fmaf(b2f((u16)uv[j]), xf, u);
-        }
-      } else {
-        for (int k = k0; k < K; ++k) {
-          g = fmaf(b2f(wg[k]), b2f(x[k]), g);
-          u = fmaf(b2f(wu[k]), b2f(x[k]), u);
+          g = fmaf(b2f((u16)gv[j]), xr[i * 8 + j], g);
+          u = fmaf(b2f((u16)uv[j]), xr[i * 8 + j], u);
         }
       }
     }
     accg[r] = g;
     accu[r] = u;
   }
-  __shared__ float redg[8][4], redu[8][4];
-  const int wid = t / WAVE, lane = t % WAVE;
 #pragma unroll
   for (int r = 0; r < 8; ++r) {
     float g = wave_sum(accg[r]);
@@ -400,15 +541,19 @@ __global__ void k_rope_store_prefill(u16* __restrict__ qkv,
 
 // ---------------------------------------------------------------------------
 // Decode attention (attention.rs:300-343 semantics, GQA, f32 softmax), over
-// the preallocated cache.  Split-KV: grid (NCHUNK, nh); each block computes
-// an online-softmax partial over a contiguous slice of positions, written to
-// ws[h][chunk] = {o[hd], m, l}; k_attn_combine reduces the partials.
+// the preallocated cache.  Split-KV in ONE launch: grid (NCHUNK, nh); each
+// block computes an online-softmax partial over a contiguous slice of
+// positions into ws[h][chunk] = {o[hd], m, l}; the LAST-arriving block of a
+// head combines the partials (agent-scope release/acquire + arrival counter
+// per cdna_hip_programming.md §6 Guideline 16 — placement-independent).
+// cnt[nh] must be zeroed before every launch (hipMemsetAsync node).
 // K/V row loads: 64 lanes x 4 B = one coalesced 256 B transaction.
 // ---------------------------------------------------------------------------
-__global__ __launch_bounds__(256) void k_attn_decode_part(
+__global__ __launch_bounds__(256) void k_attn_decode_fused(
     const u16* __restrict__ q,         // (nh*hd), post-rope
     const u16* __restrict__ kc, const u16* __restrict__ vc,
-    const int* __restrict__ pos, float* __restrict__ ws, int nh, int nkv,
+    const int* __restrict__ pos, float* __restrict__ ws,
+    u32* __restrict__ cnt, u16* __restrict__ outbuf, int nh, int nkv,
     int hd, int max_seq, int nchunk) {
   const int h = blockIdx.y;
   const int chunk = blockIdx.x;
@@ -469,47 +614,48 @@ __global__ __launch_bounds__(256) void k_attn_decode_part(
     wsrow[hd + 1] = L;
   }
   __syncthreads();
-  float M = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
+  {
+    float M = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
+    for (int d = t; d < hd; d += blockDim.x) {
+      float o = 0.f;
+#pragma unroll
+      for (int w = 0; w < 4; ++w)
+        o += (sm[w] == -INFINITY) ? 0.f : so[w][d] * __expf(sm[w] - M);
+      wsrow[d] = o;
+    }
+  }
+
+  // ---- publish partial + elect the combining block (Guideline 16 R1) ----
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // every wave drains
+  __syncthreads();
+  if (t == 0) {
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    u32 v = __hip_atomic_fetch_add(&cnt[h], 1u, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+    sm[0] = (v == (u32)(nchunk - 1)) ? 1.f : 0.f;  // reuse existing shared
+  }
+  __syncthreads();
+  if (sm[0] == 0.f) return;
+  if (t == 0) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  __syncthreads();
+
+  // ---- combine this head's partials (runs in exactly one block) ----------
+  const float* base = ws + (size_t)h * nchunk * (hd + 2);
+  float M = -INFINITY;
+  for (int c = 0; c < nchunk; ++c) M = fmaxf(M, base[c * (hd + 2) + hd]);
+  float L = 0.f;
+  for (int c = 0; c < nchunk; ++c) {
+    float mc = base[c * (hd + 2) + hd];
+    if (mc != -INFINITY) L += base[c * (hd + 2) + hd + 1] * __expf(mc - M);
+  }
   for (int d = t; d < hd; d += blockDim.x) {
     float o = 0.f;
-#pragma unroll
-    for (int w = 0; w < 4; ++w)
-      o += (sm[w] == -INFINITY) ? 0.f : so[w][d] * __expf(sm[w] - M);
-    wsrow[d] = o;
-  }
-}
-
-__global__ void k_attn_decode_combine(const float* __restrict__ ws,
-                                      u16* __restrict__ out, int nh, int hd,
-                                      int nchunk) {
-  const int h = blockIdx.x;
-  const float* base = ws + (size_t)h * nchunk * (hd + 2);
-  __shared__ float M;
-  if (threadIdx.x == 0) {
-    float m = -INFINITY;
-    for (int c = 0; c < nchunk; ++c) m = fmaxf(m, base[c * (hd + 2) + hd]);
-    M = m;
-  }
-  __syncthreads();
-  float Mv = M;
-  __shared__ float L;
-  if (threadIdx.x == 0) {
-    float l = 0.f;
     for (int c = 0; c < nchunk; ++c) {
       float mc = base[c * (hd + 2) + hd];
-      if (mc != -INFINITY) l += base[c * (hd + 2) + hd + 1] * __expf(mc - Mv);
+      if (mc != -INFINITY) o += base[c * (hd + 2) + d] * __expf(mc - M);
     }
-    L = l;
-  }
-  __syncthreads();
-  float Lv = L;
-  for (int d = threadIdx.x; d < hd; d += blockDim.x) {
-    float o = 0.f;
-    for (int c = 0; c < nchunk; ++c) {
-      float mc = base[c * (hd + 2) + hd];
-      if (mc != -INFINITY) o += base[c * (hd + 2) + d] * __expf(mc - Mv);
-    }
-    out[(size_t)h * hd + d] = f2b(o / Lv);
+    outbuf[(size_t)h * hd + d] = f2b(o / L);
   }
 }
 
@@ -830,20 +976,69 @@ void launch_silu_mul(const u16* g, const u16* u, u16* out, size_t n,
   int blocks = (int)min((n + 255) / 256, (size_t)4096);
   hipLaunchKernelGGL(k_silu_mul, dim3(blocks), dim3(256), 0, s, g, u, out, n);
 }
-void launch_gemv(const u16* W, const u16* x, void* out, const u16* res, int N,
-                 int K, int epi, hipStream_t s) {
-  dim3 grid((N + 7) / 8);
-  if (epi == 0)
-    hipLaunchKernelGGL(k_gemv<0>, grid, dim3(256), 0, s, W, x, out, res, N, K);
-  else if (epi == 1)
-    hipLaunchKernelGGL(k_gemv<1>, grid, dim3(256), 0, s, W, x, out, res, N, K);
-  else
-    hipLaunchKernelGGL(k_gemv<2>, grid, dim3(256), 0, s, W, x, out, res, N, K);
+template <int ROWS, int EPI>
+static void gemv_dispatch_kb(const u16* W, const u16* x, void* out,
+                             const u16* res, const u16* nw, float eps, int N,
+                             int K, hipStream_t s) {
+  dim3 grid((N + ROWS - 1) / ROWS);
+  if (K > 16384) {
+    hipLaunchKernelGGL((k_gemv_stream<ROWS, EPI>), grid, dim3(256), 0, s, W,
+                       x, out, res, N, K);
+    return;
+  }
+#define GEMV_KB(KB)                                                        \
+  do {                                                                     \
+    if (nw)                                                                \
+      hipLaunchKernelGGL((k_gemv_reg<ROWS, EPI, true, KB>), grid,          \
+                         dim3(256), 0, s, W, x, out, res, nw, eps, N, K);  \
+    else                                                                   \
+      hipLaunchKernelGGL((k_gemv_reg<ROWS, EPI, false, KB>), grid,         \
+                         dim3(256), 0, s, W, x, out, res, nw, eps, N, K);  \
+  } while (0)
+  if (K <= 4096) GEMV_KB(2);
+  else if (K <= 8192) GEMV_KB(4);
+  else GEMV_KB(8);
+#undef GEMV_KB
 }
-void launch_gemv_gateup(const u16* W, const u16* x, u16* out, int I, int K,
-                        hipStream_t s) {
+
+void launch_gemv(const u16* W, const u16* x, void* out, const u16* res,
+                 const u16* nw, float eps, int N, int K, int epi,
+                 hipStream_t s) {
+  // small row-count per block keeps the grid >= ~2048 workgroups on the
+  // usual decode shapes (256 CUs need many blocks to reach HBM peak)
+  const bool big = N >= 16384;
+  if (epi == 0) {
+    if (big) gemv_dispatch_kb<8, 0>(W, x, out, res, nw, eps, N, K, s);
+    else gemv_dispatch_kb<2, 0>(W, x, out, res, nw, eps, N, K, s);
+  } else if (epi == 1) {
+    if (big) gemv_dispatch_kb<8, 1>(W, x, out, res, nw, eps, N, K, s);
+    else gemv_dispatch_kb<2, 1>(W, x, out, res, nw, eps, N, K, s);
+  } else {
+    if (big) gemv_dispatch_kb<8, 2>(W, x, out, res, nw, eps, N, K, s);
+    else gemv_dispatch_kb<2, 2>(W, x, out, res, nw, eps, N, K, s);
+  }
+}
+void launch_gemv_gateup(const u16* W, const u16* x, u16* out, const u16* nw,
+                        float eps, int I, int K, hipStream_t s) {
   dim3 grid((I + 7) / 8);
-  hipLaunchKernelGGL(k_gemv_gateup, grid, dim3(256), 0, s, W, x, out, I, K);
+  if (K > 16384) {
+    // not on any config's gate_up path; fall back via two stream GEMVs is
+    // unnecessary — guard loudly at engine level (K = hidden <= 16384)
+    return;
+  }
+#define GU_KB(KB)                                                          \
+  do {                                                                     \
+    if (nw)                                                                \
+      hipLaunchKernelGGL((k_gemv_gateup<true, KB>), grid, dim3(256), 0, s, \
+                         W, x, out, nw, eps, I, K);                        \
+    else                                                                   \
+      hipLaunchKernelGGL((k_gemv_gateup<false, KB>), grid, dim3(256), 0,   \
+                         s, W, x, out, nw, eps, I, K);                     \
+  } while (0)
+  if (K <= 4096) GU_KB(2);
+  else if (K <= 8192) GU_KB(4);
+  else GU_KB(8);
+#undef GU_KB
 }
 void launch_embed_token(const u16* embed, const u32* tok, u16* x, int H,
                         hipStream_t s) {
@@ -869,12 +1064,15 @@ void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, const float* cost,
                      max_seq, qkv_stride);
 }
 void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
-                        const int* pos, float* ws, u16* out, int nh, int nkv,
-                        int hd, int max_seq, int nchunk, hipStream_t s) {
-  hipLaunchKernelGGL(k_attn_decode_part, dim3(nchunk, nh), dim3(256), 0, s, q,
-                     kc, vc, pos, ws, nh, nkv, hd, max_seq, nchunk);
-  hipLaunchKernelGGL(k_attn_decode_combine, dim3(nh), dim3(128), 0, s, ws, out,
-                     nh, hd, nchunk);
+                        const int* pos, float* ws, u32* cnt, u16* out, int nh,
+                        int nkv, int hd, int max_seq, int nchunk,
+                        hipStream_t s) {
+  // arrival counters must be zero at launch (Guideline 16: re-initialise
+  // every call via a memset node — graph-capturable)
+  hipMemsetAsync(cnt, 0, sizeof(u32) * nh, s);
+  hipLaunchKernelGGL(k_attn_decode_fused, dim3(nchunk, nh), dim3(256), 0, s,
+                     q, kc, vc, pos, ws, cnt, out, nh, nkv, hd, max_seq,
+                     nchunk);
 }
 void launch_attn_prefill(const u16* qkv, const u16* kc, const u16* vc,
                          u16* out, int S, int pos0, int nh, int nkv, int hd,
