@@ -51,9 +51,10 @@ __device__ __host__ inline float b2f(u16 h) {
 
 // nontemporal 16-B weight load: decode weights are streamed exactly once per
 // token — keep them from thrashing L2/L3 (guide G14: streaming data)
+using uint4v = __attribute__((ext_vector_type(4))) unsigned int;
 __device__ inline short8 ntload8(const u16* p) {
-  uint4 v = __builtin_nontemporal_load(reinterpret_cast<const uint4*>(p));
-  union { uint4 u; short8 s; } c{v};
+  uint4v v = __builtin_nontemporal_load(reinterpret_cast<const uint4v*>(p));
+  union { uint4v u; short8 s; } c{v};
   return c.s;
 }
 
