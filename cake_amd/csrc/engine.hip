@@ -229,7 +229,8 @@ struct cake_engine {
   u32* dev_tok = nullptr;
   int host_pos = 0;
 
-  int nchunk = 8;  // split-KV chunks for decode attention (CAKE_NCHUNK)
+  int nchunk = 8;   // split-KV chunks for decode attention (CAKE_NCHUNK)
+  int gu_rows = 4;  // gate_up channels per block (CAKE_GU_ROWS)
 
   hipStream_t stream = nullptr;
   hipGraphExec_t graph = nullptr;
@@ -354,7 +355,7 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
     StatScope ss(e, "gemv_gateup", 2.0 * I * H * 2 + 2.0 * H * 2 + I * 2,
                  4.0 * I * H);
     launch_gemv_gateup(l.wgu, e->x, e->act, l.rms2, c.rms_eps, I, H,
-                       e->stream);
+                       e->gu_rows, e->stream);
   }
   {  // down projection + residual
     StatScope ss(e, "gemv_down", (double)H * I * 2 + I * 2 + H * 4,
@@ -558,8 +559,13 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
     int v = atoi(nc);
     if (v >= 1 && v <= 64) e->nchunk = v;
   }
+  if (const char* gr = getenv("CAKE_GU_ROWS")) {
+    int v = atoi(gr);
+    if (v == 4 || v == 8) e->gu_rows = v;
+  }
   ALLOC(e->attn_ws, float, (size_t)c.nh * e->nchunk * (hd + 2));
   ALLOC(e->attn_cnt, u32, c.nh);
+  HIP_TRY(hipMemset(e->attn_cnt, 0, sizeof(u32) * c.nh));
   ALLOC(e->dev_pos, int, 1);
   ALLOC(e->dev_step, int, 1);
   ALLOC(e->dev_tok, u32, 1);

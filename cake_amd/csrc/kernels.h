@@ -25,7 +25,7 @@ void launch_gemv(const u16* W, const u16* x, void* out, const u16* res,
                  const u16* nw, float eps, int N, int K, int epi,
                  hipStream_t s);
 void launch_gemv_gateup(const u16* W, const u16* x, u16* out, const u16* nw,
-                        float eps, int I, int K, hipStream_t s);
+                        float eps, int I, int K, int rows, hipStream_t s);
 void launch_embed_token(const u16* embed, const u32* tok, u16* x, int H,
                         hipStream_t s);
 void launch_embed_rows(const u16* embed, const u32* ids, u16* x, int S, int H,

@@ -353,13 +353,13 @@ __global__ __launch_bounds__(256) void k_gemv_stream(
 // fused [rms_norm ->] gate_up GEMV -> silu_mul: block computes
 // out[i] = silu(g_i) * u_i for 8 channels, g_i = dot(W[i,:], xn),
 // u_i = dot(W[i+I,:], xn)  (mlp.rs:21-31 + fused gate_up of mlp.rs:38-46)
-template <bool NORM, int KB>
+template <int ROWS, bool NORM, int KB>
 __global__ __launch_bounds__(256) void k_gemv_gateup(
     const u16* __restrict__ W, const u16* __restrict__ x,
     u16* __restrict__ out, const u16* __restrict__ nw, float eps, int I,
     int K) {
   const int t = threadIdx.x;
-  const int c0 = blockIdx.x * 8;
+  const int c0 = blockIdx.x * ROWS;
   const int wid = t / WAVE, lane = t % WAVE;
   __shared__ float redg[8][4], redu[8][4];
 
@@ -399,11 +399,11 @@ __global__ __launch_bounds__(256) void k_gemv_gateup(
     }
   }
 
-  float accg[8], accu[8];
+  float accg[ROWS], accu[ROWS];
 #pragma unroll
-  for (int r = 0; r < 8; ++r) accg[r] = accu[r] = 0.f;
+  for (int r = 0; r < ROWS; ++r) accg[r] = accu[r] = 0.f;
 #pragma unroll 1
-  for (int r = 0; r < 8; ++r) {
+  for (int r = 0; r < ROWS; ++r) {
     const int c = c0 + r;
     if (c >= I) break;
     const u16* wg = W + (size_t)c * K;
@@ -426,13 +426,13 @@ __global__ __launch_bounds__(256) void k_gemv_gateup(
     accu[r] = u;
   }
 #pragma unroll
-  for (int r = 0; r < 8; ++r) {
+  for (int r = 0; r < ROWS; ++r) {
     float g = wave_sum(accg[r]);
     float u = wave_sum(accu[r]);
     if (lane == 0) { redg[r][wid] = g; redu[r][wid] = u; }
   }
   __syncthreads();
-  if (t < 8 && c0 + t < I) {
+  if (t < ROWS && c0 + t < I) {
     float g = redg[t][0] + redg[t][1] + redg[t][2] + redg[t][3];
     float u = redu[t][0] + redu[t][1] + redu[t][2] + redu[t][3];
     out[c0 + t] = f2b(g / (1.f + __expf(-g)) * u);
@@ -632,9 +632,11 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
   if (t == 0) {
     __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    // epoch-free election: counters monotonically accumulate, the block
+    // drawing (v % nchunk) == nchunk-1 combines — no per-launch reset
     u32 v = __hip_atomic_fetch_add(&cnt[h], 1u, __ATOMIC_RELAXED,
                                    __HIP_MEMORY_SCOPE_AGENT);
-    sm[0] = (v == (u32)(nchunk - 1)) ? 1.f : 0.f;  // reuse existing shared
+    sm[0] = (v % (u32)nchunk == (u32)(nchunk - 1)) ? 1.f : 0.f;
   }
   __syncthreads();
   if (sm[0] == 0.f) return;
@@ -1007,39 +1009,47 @@ void launch_gemv(const u16* W, const u16* x, void* out, const u16* res,
                  hipStream_t s) {
   // small row-count per block keeps the grid >= ~2048 workgroups on the
   // usual decode shapes (256 CUs need many blocks to reach HBM peak)
-  const bool big = N >= 16384;
+  const int rows = N >= 16384 ? 8 : (N >= 4096 ? 4 : 2);
   if (epi == 0) {
-    if (big) gemv_dispatch_kb<8, 0>(W, x, out, res, nw, eps, N, K, s);
+    if (rows == 8) gemv_dispatch_kb<8, 0>(W, x, out, res, nw, eps, N, K, s);
+    else if (rows == 4)
+      gemv_dispatch_kb<4, 0>(W, x, out, res, nw, eps, N, K, s);
     else gemv_dispatch_kb<2, 0>(W, x, out, res, nw, eps, N, K, s);
   } else if (epi == 1) {
-    if (big) gemv_dispatch_kb<8, 1>(W, x, out, res, nw, eps, N, K, s);
+    if (rows == 8) gemv_dispatch_kb<8, 1>(W, x, out, res, nw, eps, N, K, s);
+    else if (rows == 4)
+      gemv_dispatch_kb<4, 1>(W, x, out, res, nw, eps, N, K, s);
     else gemv_dispatch_kb<2, 1>(W, x, out, res, nw, eps, N, K, s);
   } else {
-    if (big) gemv_dispatch_kb<8, 2>(W, x, out, res, nw, eps, N, K, s);
+    if (rows == 8) gemv_dispatch_kb<8, 2>(W, x, out, res, nw, eps, N, K, s);
+    else if (rows == 4)
+      gemv_dispatch_kb<4, 2>(W, x, out, res, nw, eps, N, K, s);
     else gemv_dispatch_kb<2, 2>(W, x, out, res, nw, eps, N, K, s);
   }
 }
 void launch_gemv_gateup(const u16* W, const u16* x, u16* out, const u16* nw,
-                        float eps, int I, int K, hipStream_t s) {
-  dim3 grid((I + 7) / 8);
-  if (K > 16384) {
-    // not on any config's gate_up path; fall back via two stream GEMVs is
-    // unnecessary — guard loudly at engine level (K = hidden <= 16384)
-    return;
-  }
-#define GU_KB(KB)                                                          \
-  do {                                                                     \
-    if (nw)                                                                \
-      hipLaunchKernelGGL((k_gemv_gateup<true, KB>), grid, dim3(256), 0, s, \
-                         W, x, out, nw, eps, I, K);                        \
-    else                                                                   \
-      hipLaunchKernelGGL((k_gemv_gateup<false, KB>), grid, dim3(256), 0,   \
-                         s, W, x, out, nw, eps, I, K);                     \
+                        float eps, int I, int K, int rows, hipStream_t s) {
+  if (K > 16384) return;  // guarded at engine create (K = hidden <= 16384)
+#define GU_KB2(ROWS, KB)                                                    \
+  do {                                                                      \
+    dim3 grid((I + ROWS - 1) / ROWS);                                       \
+    if (nw)                                                                 \
+      hipLaunchKernelGGL((k_gemv_gateup<ROWS, true, KB>), grid, dim3(256),  \
+                         0, s, W, x, out, nw, eps, I, K);                   \
+    else                                                                    \
+      hipLaunchKernelGGL((k_gemv_gateup<ROWS, false, KB>), grid, dim3(256), \
+                         0, s, W, x, out, nw, eps, I, K);                   \
+  } while (0)
+#define GU_KB(KB)                                                           \
+  do {                                                                      \
+    if (rows >= 8) GU_KB2(8, KB);                                           \
+    else GU_KB2(4, KB);                                                     \
   } while (0)
   if (K <= 4096) GU_KB(2);
   else if (K <= 8192) GU_KB(4);
   else GU_KB(8);
 #undef GU_KB
+#undef GU_KB2
 }
 void launch_embed_token(const u16* embed, const u32* tok, u16* x, int H,
                         hipStream_t s) {
@@ -1068,9 +1078,6 @@ void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
                         const int* pos, float* ws, u32* cnt, u16* out, int nh,
                         int nkv, int hd, int max_seq, int nchunk,
                         hipStream_t s) {
-  // arrival counters must be zero at launch (Guideline 16: re-initialise
-  // every call via a memset node — graph-capturable)
-  hipMemsetAsync(cnt, 0, sizeof(u32) * nh, s);
   hipLaunchKernelGGL(k_attn_decode_fused, dim3(nchunk, nh), dim3(256), 0, s,
                      q, kc, vc, pos, ws, cnt, out, nh, nkv, hd, max_seq,
                      nchunk);
