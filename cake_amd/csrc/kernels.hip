@@ -26,6 +26,7 @@
 #include <hip/hip_bf16.h>
 
 #include <cstdint>
+#include <cstdlib>
 
 #define WAVE 64
 
@@ -1009,23 +1010,28 @@ void launch_gemv(const u16* W, const u16* x, void* out, const u16* res,
                  hipStream_t s) {
   // small row-count per block keeps the grid >= ~2048 workgroups on the
   // usual decode shapes (256 CUs need many blocks to reach HBM peak)
-  const int rows = N >= 16384 ? 8 : (N >= 4096 ? 4 : 2);
-  if (epi == 0) {
-    if (rows == 8) gemv_dispatch_kb<8, 0>(W, x, out, res, nw, eps, N, K, s);
-    else if (rows == 4)
-      gemv_dispatch_kb<4, 0>(W, x, out, res, nw, eps, N, K, s);
-    else gemv_dispatch_kb<2, 0>(W, x, out, res, nw, eps, N, K, s);
-  } else if (epi == 1) {
-    if (rows == 8) gemv_dispatch_kb<8, 1>(W, x, out, res, nw, eps, N, K, s);
-    else if (rows == 4)
-      gemv_dispatch_kb<4, 1>(W, x, out, res, nw, eps, N, K, s);
-    else gemv_dispatch_kb<2, 1>(W, x, out, res, nw, eps, N, K, s);
-  } else {
-    if (rows == 8) gemv_dispatch_kb<8, 2>(W, x, out, res, nw, eps, N, K, s);
-    else if (rows == 4)
-      gemv_dispatch_kb<4, 2>(W, x, out, res, nw, eps, N, K, s);
-    else gemv_dispatch_kb<2, 2>(W, x, out, res, nw, eps, N, K, s);
-  }
+  static const int env_small = [] {
+    const char* v = getenv("CAKE_GEMV_ROWS_SMALL");
+    return v ? atoi(v) : 0;
+  }();
+  int rows;
+  if (N >= 16384) rows = 8;                 // lm_head: maximal block count
+  else if (env_small) rows = env_small;     // A/B override
+  else if (K >= 8192) rows = 4;             // long rows amortize x reload
+  else rows = (N >= 6144) ? 4 : 2;
+#define GEMV_R(R, EPI) gemv_dispatch_kb<R, EPI>(W, x, out, res, nw, eps, N, K, s)
+#define GEMV_EPI(EPI)                          \
+  do {                                         \
+    if (rows >= 8) GEMV_R(8, EPI);             \
+    else if (rows == 4) GEMV_R(4, EPI);        \
+    else if (rows == 2) GEMV_R(2, EPI);        \
+    else GEMV_R(1, EPI);                       \
+  } while (0)
+  if (epi == 0) GEMV_EPI(0);
+  else if (epi == 1) GEMV_EPI(1);
+  else GEMV_EPI(2);
+#undef GEMV_EPI
+#undef GEMV_R
 }
 void launch_gemv_gateup(const u16* W, const u16* x, u16* out, const u16* nw,
                         float eps, int I, int K, int rows, hipStream_t s) {
