@@ -403,28 +403,27 @@ __global__ __launch_bounds__(256) void k_gemv_gateup(
   float accg[ROWS], accu[ROWS];
 #pragma unroll
   for (int r = 0; r < ROWS; ++r) accg[r] = accu[r] = 0.f;
-#pragma unroll 1
-  for (int r = 0; r < ROWS; ++r) {
-    const int c = c0 + r;
-    if (c >= I) break;
-    const u16* wg = W + (size_t)c * K;
-    const u16* wu = W + (size_t)(c + I) * K;
-    float g = 0.f, u = 0.f;
 #pragma unroll
-    for (int i = 0; i < KB; ++i) {
-      const int k0 = i * 2048 + t * 8;
-      if (k0 < K) {
-        short8 gv = ntload8(wg + k0);
-        short8 uv = ntload8(wu + k0);
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    if (k0 < K) {
+      short8 gv[ROWS], uv[ROWS];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          g = fmaf(b2f((u16)gv[j]), xr[i * 8 + j], g);
-          u = fmaf(b2f((u16)uv[j]), xr[i * 8 + j], u);
+      for (int r = 0; r < ROWS; ++r)
+        if (c0 + r < I) {
+          gv[r] = ntload8(W + (size_t)(c0 + r) * K + k0);
+          uv[r] = ntload8(W + (size_t)(c0 + r + I) * K + k0);
         }
-      }
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r)
+        if (c0 + r < I) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            accg[r] = fmaf(b2f((u16)gv[r][j]), xr[i * 8 + j], accg[r]);
+            accu[r] = fmaf(b2f((u16)uv[r][j]), xr[i * 8 + j], accu[r]);
+          }
+        }
     }
-    accg[r] = g;
-    accu[r] = u;
   }
 #pragma unroll
   for (int r = 0; r < ROWS; ++r) {
@@ -568,6 +567,10 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
   const int e0 = 2 * lane;             // dims (2*lane, 2*lane+1)
   const bool act = e0 + 1 < hd;
   float* wsrow = ws + ((size_t)h * nchunk + chunk) * (hd + 2);
+#define WS_STORE(p, v)                                                     \
+  __hip_atomic_store((p), (v), __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)
+#define WS_LOAD(p)                                                         \
+  __hip_atomic_load((p), __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)
 
   float q0 = 0.f, q1 = 0.f;
   if (act) {
@@ -612,8 +615,8 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
 #pragma unroll
     for (int w = 0; w < 4; ++w)
       L += (sm[w] == -INFINITY) ? 0.f : sl[w] * __expf(sm[w] - M);
-    wsrow[hd] = M;
-    wsrow[hd + 1] = L;
+    WS_STORE(&wsrow[hd], M);
+    WS_STORE(&wsrow[hd + 1], L);
   }
   __syncthreads();
   {
@@ -623,16 +626,17 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
 #pragma unroll
       for (int w = 0; w < 4; ++w)
         o += (sm[w] == -INFINITY) ? 0.f : so[w][d] * __expf(sm[w] - M);
-      wsrow[d] = o;
+      WS_STORE(&wsrow[d], o);
     }
   }
 
-  // ---- publish partial + elect the combining block (Guideline 16 R1) ----
+  // ---- publish partial + elect the combining block ----------------------
+  // sc1 write-through publish (Guideline 16 R1 variant): the ws stores above
+  // are agent-scope relaxed 4-B atomics (= sc1 stores, the natural width
+  // here), so no release fence is needed — just drain, then count arrivals.
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // every wave drains
   __syncthreads();
   if (t == 0) {
-    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     // epoch-free election: counters monotonically accumulate, the block
     // drawing (v % nchunk) == nchunk-1 combines — no per-launch reset
     u32 v = __hip_atomic_fetch_add(&cnt[h], 1u, __ATOMIC_RELAXED,
@@ -641,26 +645,30 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
   }
   __syncthreads();
   if (sm[0] == 0.f) return;
-  if (t == 0) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-  __syncthreads();
+  // reducer reads the slabs with sc1 loads — no acquire fence, no L1 risk
 
   // ---- combine this head's partials (runs in exactly one block) ----------
-  const float* base = ws + (size_t)h * nchunk * (hd + 2);
+  float* base = ws + (size_t)h * nchunk * (hd + 2);
   float M = -INFINITY;
-  for (int c = 0; c < nchunk; ++c) M = fmaxf(M, base[c * (hd + 2) + hd]);
+  for (int c = 0; c < nchunk; ++c)
+    M = fmaxf(M, WS_LOAD(&base[c * (hd + 2) + hd]));
   float L = 0.f;
   for (int c = 0; c < nchunk; ++c) {
-    float mc = base[c * (hd + 2) + hd];
-    if (mc != -INFINITY) L += base[c * (hd + 2) + hd + 1] * __expf(mc - M);
+    float mc = WS_LOAD(&base[c * (hd + 2) + hd]);
+    if (mc != -INFINITY)
+      L += WS_LOAD(&base[c * (hd + 2) + hd + 1]) * __expf(mc - M);
   }
   for (int d = t; d < hd; d += blockDim.x) {
     float o = 0.f;
     for (int c = 0; c < nchunk; ++c) {
-      float mc = base[c * (hd + 2) + hd];
-      if (mc != -INFINITY) o += base[c * (hd + 2) + d] * __expf(mc - M);
+      float mc = WS_LOAD(&base[c * (hd + 2) + hd]);
+      if (mc != -INFINITY)
+        o += WS_LOAD(&base[c * (hd + 2) + d]) * __expf(mc - M);
     }
     outbuf[(size_t)h * hd + d] = f2b(o / L);
   }
+#undef WS_STORE
+#undef WS_LOAD
 }
 
 // ---------------------------------------------------------------------------
