@@ -215,6 +215,21 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
   const int wid = t / WAVE, lane = t % WAVE;
   __shared__ float red[ROWS > 4 ? ROWS : 4][4];
 
+  // For short K, issue ALL weight loads first — they stay in flight across
+  // the x/norm phase (plain VGPR loads survive s_barrier; guide §5
+  // pipelining note), hiding the norm reduction latency entirely.
+  short8 wpre[KB <= 2 ? ROWS : 1][KB <= 2 ? KB : 1];
+  if (KB <= 2) {
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+#pragma unroll
+      for (int i = 0; i < KB; ++i) {
+        const int k0 = i * 2048 + t * 8;
+        if (row0 + r < N && k0 < K)
+          wpre[r][i] = ntload8(W + (size_t)(row0 + r) * K + k0);
+      }
+  }
+
   // phase 1: x -> registers (f32), optionally fused rms_norm
   float xr[KB * 8];
 #pragma unroll
@@ -253,27 +268,41 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
     }
   }
 
-  // phase 2: stream ROWS weight rows
+  // phase 2: FMA with the prefetched weights (KB<=2) or stream rows with
+  // all rows' loads unrolled together (KB>2) — memory-level parallelism
   float acc[ROWS];
 #pragma unroll
   for (int r = 0; r < ROWS; ++r) acc[r] = 0.f;
-#pragma unroll 1
-  for (int r = 0; r < ROWS; ++r) {
-    const int row = row0 + r;
-    if (row >= N) break;
-    const u16* wr = W + (size_t)row * K;
-    float a = 0.f;
+  if (KB <= 2) {
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+#pragma unroll
+      for (int i = 0; i < KB; ++i) {
+        const int k0 = i * 2048 + t * 8;
+        if (row0 + r < N && k0 < K) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            acc[r] = fmaf(b2f((u16)wpre[r][i][j]), xr[i * 8 + j], acc[r]);
+        }
+      }
+  } else {
 #pragma unroll
     for (int i = 0; i < KB; ++i) {
       const int k0 = i * 2048 + t * 8;
       if (k0 < K) {
-        short8 wv = *reinterpret_cast<const short8*>(wr + k0);
+        short8 wv[ROWS];
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          a = fmaf(b2f((u16)wv[j]), xr[i * 8 + j], a);
+        for (int r = 0; r < ROWS; ++r)
+          if (row0 + r < N) wv[r] = ntload8(W + (size_t)(row0 + r) * K + k0);
+#pragma unroll
+        for (int r = 0; r < ROWS; ++r)
+          if (row0 + r < N) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              acc[r] = fmaf(b2f((u16)wv[r][j]), xr[i * 8 + j], acc[r]);
+          }
       }
     }
-    acc[r] = a;
   }
 #pragma unroll
   for (int r = 0; r < ROWS; ++r) {
@@ -446,7 +475,11 @@ __global__ void k_embed_token(const u16* __restrict__ embed,
                               const u32* __restrict__ tok,
                               u16* __restrict__ x, int H) {
   const u16* src = embed + (size_t)(*tok) * H;
-  for (int i = threadIdx.x; i < H; i += blockDim.x) x[i] = src[i];
+  const int nv = H / 8;
+  for (int i = threadIdx.x; i < nv; i += blockDim.x)
+    *reinterpret_cast<short8*>(x + i * 8) =
+        *reinterpret_cast<const short8*>(src + i * 8);
+  for (int i = nv * 8 + threadIdx.x; i < H; i += blockDim.x) x[i] = src[i];
 }
 __global__ void k_embed_rows(const u16* __restrict__ embed,
                              const u32* __restrict__ ids,
@@ -649,22 +682,25 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
 
   // ---- combine this head's partials (runs in exactly one block) ----------
   float* base = ws + (size_t)h * nchunk * (hd + 2);
-  float M = -INFINITY;
-  for (int c = 0; c < nchunk; ++c)
-    M = fmaxf(M, WS_LOAD(&base[c * (hd + 2) + hd]));
-  float L = 0.f;
-  for (int c = 0; c < nchunk; ++c) {
-    float mc = WS_LOAD(&base[c * (hd + 2) + hd]);
-    if (mc != -INFINITY)
-      L += WS_LOAD(&base[c * (hd + 2) + hd + 1]) * __expf(mc - M);
+  // stage m,l in LDS (parallel sc1 loads; serial dependent uncached loads
+  // were the reducer's cost), then combine with per-chunk weights from LDS
+  if (t < nchunk) {
+    sm[0] = 0.f;  // keep sm[0] clear; use so rows as staging
+    so[0][t] = WS_LOAD(&base[t * (hd + 2) + hd]);
+    so[1][t] = WS_LOAD(&base[t * (hd + 2) + hd + 1]);
   }
+  __syncthreads();
+  float M = -INFINITY;
+  for (int c = 0; c < nchunk; ++c) M = fmaxf(M, so[0][c]);
+  float L = 0.f;
+  for (int c = 0; c < nchunk; ++c)
+    if (so[0][c] != -INFINITY) L += so[1][c] * __expf(so[0][c] - M);
   for (int d = t; d < hd; d += blockDim.x) {
     float o = 0.f;
-    for (int c = 0; c < nchunk; ++c) {
-      float mc = WS_LOAD(&base[c * (hd + 2) + hd]);
-      if (mc != -INFINITY)
-        o += WS_LOAD(&base[c * (hd + 2) + d]) * __expf(mc - M);
-    }
+#pragma unroll 4
+    for (int c = 0; c < nchunk; ++c)
+      if (so[0][c] != -INFINITY)
+        o += WS_LOAD(&base[c * (hd + 2) + d]) * __expf(so[0][c] - M);
     outbuf[(size_t)h * hd + d] = f2b(o / L);
   }
 #undef WS_STORE
