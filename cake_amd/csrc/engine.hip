@@ -197,6 +197,7 @@ struct LayerDev {
   u16 *wqkv = nullptr, *wo = nullptr, *wgu = nullptr, *wdown = nullptr;
   u16 *qnorm = nullptr, *knorm = nullptr;
   u16 *kc = nullptr, *vc = nullptr;  // (nkv, max_seq, hd) each
+  u16 *vtc = nullptr;                // V transposed: (nkv, hd, max_seq)
 };
 
 struct cake_engine {
@@ -333,7 +334,7 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
   }
   {  // rope q,k + KV store at slot *pos
     StatScope ss(e, "rope_store", (double)Nq * hd * 0, 0);
-    launch_rope_store_decode(e->qkv, l.kc, l.vc, e->cos_t, e->sin_t,
+    launch_rope_store_decode(e->qkv, l.kc, l.vc, l.vtc, e->cos_t, e->sin_t,
                              e->dev_pos, c.nh, c.nkv, hd, hd, e->max_seq,
                              e->stream);
   }
@@ -388,15 +389,16 @@ static void enqueue_layer_prefill(cake_engine* e, LayerDev& l, int S,
   }
   {
     StatScope ss(e, "rope_store_pf", 0, 0);
-    launch_rope_store_prefill(e->qkv, l.kc, l.vc, e->cos_t, e->sin_t, pos0, S,
-                              c.nh, c.nkv, hd, hd, e->max_seq, Nq, e->stream);
+    launch_rope_store_prefill(e->qkv, l.kc, l.vc, l.vtc, e->cos_t, e->sin_t,
+                              pos0, S, c.nh, c.nkv, hd, hd, e->max_seq, Nq,
+                              e->stream);
   }
   {
     double n_avg = pos0 + (S + 1) * 0.5;
     StatScope ss(e, "attn_prefill", 2.0 * S * n_avg * 2 * hd * c.nh / 4,
                  4.0 * S * n_avg * hd * c.nh);
-    launch_attn_prefill(e->qkv, l.kc, l.vc, e->attn_out, S, pos0, c.nh, c.nkv,
-                        hd, e->max_seq, Nq, Sq, e->stream);
+    launch_attn_prefill(e->qkv, l.kc, l.vc, l.vtc, e->attn_out, S, pos0,
+                        c.nh, c.nkv, hd, e->max_seq, Nq, Sq, e->stream);
   }
   {
     StatScope ss(e, "gemm_o", (double)H * Sq * 2 + (double)S * (Sq + H) * 2,
@@ -490,6 +492,7 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
   if (c.hidden > 16384)
     return set_err(5, "hidden_size %d unsupported (> 16384)", c.hidden);
   if (max_seq <= 0) max_seq = c.max_pos;
+  max_seq = (max_seq + 31) & ~31;  // MFMA prefill reads whole 32-pos tiles
   if (max_batch_tokens <= 0) max_batch_tokens = 2048;
   if (max_batch_tokens > max_seq) max_batch_tokens = max_seq;
 
@@ -522,6 +525,7 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
     }
     ALLOC(l.kc, u16, (size_t)c.nkv * max_seq * hd);
     ALLOC(l.vc, u16, (size_t)c.nkv * max_seq * hd);
+    ALLOC(l.vtc, u16, (size_t)c.nkv * hd * max_seq);
   }
   if (e->has_embed()) ALLOC(e->embed, u16, (size_t)V * H);
   if (e->has_head()) {
@@ -587,6 +591,7 @@ extern "C" void cake_hip_engine_free(cake_engine* e) {
   for (auto& l : e->L) {
     hipFree(l.rms1); hipFree(l.rms2); hipFree(l.wqkv); hipFree(l.wo);
     hipFree(l.wgu); hipFree(l.wdown); hipFree(l.kc); hipFree(l.vc);
+    hipFree(l.vtc);
     if (l.qnorm) { hipFree(l.qnorm); hipFree(l.knorm); }
   }
   if (e->embed) hipFree(e->embed);

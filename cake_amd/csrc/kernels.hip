@@ -499,6 +499,7 @@ __global__ void k_embed_rows(const u16* __restrict__ embed,
 // ---------------------------------------------------------------------------
 __global__ void k_rope_store_decode(u16* __restrict__ qkv,
                                     u16* __restrict__ kc, u16* __restrict__ vc,
+                                    u16* __restrict__ vtc,
                                     const float* __restrict__ cost,
                                     const float* __restrict__ sint,
                                     const int* __restrict__ pos, int nh,
@@ -525,11 +526,15 @@ __global__ void k_rope_store_decode(u16* __restrict__ qkv,
       dst[i + half] = f2b(x2 * c[i] + x1 * s[i]);
     }
     for (int i = rd + threadIdx.x; i < hd; i += blockDim.x) dst[i] = k[i];
-  } else {                            // v head -> cache slot p
+  } else {                            // v head -> cache slot p (+ V^T)
     const int h = b - nh - nkv;
     const u16* v = qkv + (size_t)(nh + nkv + h) * hd;
     u16* dst = vc + ((size_t)h * max_seq + p) * hd;
-    for (int i = threadIdx.x; i < hd; i += blockDim.x) dst[i] = v[i];
+    u16* dstt = vtc + (size_t)h * hd * max_seq + p;
+    for (int i = threadIdx.x; i < hd; i += blockDim.x) {
+      dst[i] = v[i];
+      dstt[(size_t)i * max_seq] = v[i];
+    }
   }
 }
 
@@ -537,6 +542,7 @@ __global__ void k_rope_store_decode(u16* __restrict__ qkv,
 // grid = (nh + 2*nkv, S)
 __global__ void k_rope_store_prefill(u16* __restrict__ qkv,
                                      u16* __restrict__ kc, u16* __restrict__ vc,
+                                     u16* __restrict__ vtc,
                                      const float* __restrict__ cost,
                                      const float* __restrict__ sint, int pos0,
                                      int nh, int nkv, int hd, int rd,
@@ -569,7 +575,11 @@ __global__ void k_rope_store_prefill(u16* __restrict__ qkv,
     const int h = b - nh - nkv;
     const u16* v = row + (size_t)(nh + nkv + h) * hd;
     u16* dst = vc + ((size_t)h * max_seq + p) * hd;
-    for (int i = threadIdx.x; i < hd; i += blockDim.x) dst[i] = v[i];
+    u16* dstt = vtc + (size_t)h * hd * max_seq + p;
+    for (int i = threadIdx.x; i < hd; i += blockDim.x) {
+      dst[i] = v[i];
+      dstt[(size_t)i * max_seq] = v[i];
+    }
   }
 }
 
@@ -779,6 +789,165 @@ __global__ void k_rope_simple(u16* __restrict__ x,
     float x1 = b2f(xr[i]), x2 = b2f(xr[i + half]);
     xr[i] = f2b(x1 * c[i] - x2 * sn[i]);
     xr[i + half] = f2b(x2 * c[i] + x1 * sn[i]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// MFMA flash-attention prefill (hd == 128) — causal, GQA, f32 online
+// softmax, bf16 I/O.  Structure after the guide's 8-wave 32x32 ladder
+// (cdna_hip_programming.md §B "Fused attention prefill"), basic variant:
+//   - 8 waves per workgroup, each wave owns 32 query rows of one head
+//   - swapped QK^T: P = mfma(K_tile, Q_tile) so the softmax column is
+//     lane-local (col j = q = lane&31); 32x32x16 bf16 MFMA, 8 per tile
+//   - P -> bf16 pack + __shfl_xor(32) half-exchange assembles the PV
+//     A-fragments in-register (the T12 idea without inline asm)
+//   - PV reads V from a TRANSPOSED cache (vt[kvh][d][pos]) so the
+//     B-fragment's 8-consecutive-k elements are one 16-B load
+// Per 32-kv tile per wave: 16 MFMAs + ~16 16-B global loads; no LDS, no
+// barriers (waves fully independent).
+// ---------------------------------------------------------------------------
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+__global__ __launch_bounds__(512) void k_attn_prefill_mfma(
+    const u16* __restrict__ qkv, const u16* __restrict__ kc,
+    const u16* __restrict__ vtc, u16* __restrict__ out, int S, int pos0,
+    int nh, int nkv, int max_seq, int qkv_stride, int out_stride) {
+  const int hd = 128;
+  const int w = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int lhalf = lane >> 5, lq = lane & 31;
+  const int h = blockIdx.y;
+  const int kvh = h / (nh / nkv);
+  const int qb = blockIdx.x * 256 + w * 32;
+  if (qb >= S) return;
+
+  // Q fragments (persistent): B[k][j=q], lane holds q = lq, dims
+  // kk*16 + lhalf*8 .. +8
+  bf16x8 qf[8];
+  {
+    const int row = min(qb + lq, S - 1);
+    const u16* qrow = qkv + (size_t)row * qkv_stride + (size_t)h * hd;
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk)
+      qf[kk] = *reinterpret_cast<const bf16x8*>(qrow + kk * 16 + lhalf * 8);
+  }
+
+  f32x16 oacc[4];
+#pragma unroll
+  for (int db = 0; db < 4; ++db)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) oacc[db][r] = 0.f;
+  float m = -INFINITY, l = 0.f;
+  const float scale = rsqrtf((float)hd);
+  const int q_abs = pos0 + qb + lq;
+  const bool q_valid = qb + lq < S;
+  const int n_wave = pos0 + min(qb + 32, S);  // kv needed by this wave
+  const int ntiles = (n_wave + 31) / 32;
+  const u16* kbase = kc + (size_t)kvh * max_seq * hd;
+  const u16* vtbase = vtc + (size_t)kvh * hd * max_seq;
+
+  for (int tile = 0; tile < ntiles; ++tile) {
+    const int pkv = tile * 32;
+    // K fragments: A[i=kv][k], lane holds kv = lq
+    bf16x8 kf[8];
+    {
+      const u16* krow = kbase + (size_t)(pkv + lq) * hd;
+#pragma unroll
+      for (int kk = 0; kk < 8; ++kk)
+        kf[kk] =
+            *reinterpret_cast<const bf16x8*>(krow + kk * 16 + lhalf * 8);
+    }
+    f32x16 p;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) p[r] = 0.f;
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk)
+      p = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[kk], qf[kk], p, 0, 0, 0);
+    // scale + causal mask (reg r -> kv row (r&3)+8*(r>>2)+4*lhalf)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kv_abs = pkv + (r & 3) + 8 * (r >> 2) + 4 * lhalf;
+      p[r] = (q_valid && kv_abs <= q_abs) ? p[r] * scale : -INFINITY;
+    }
+    // online softmax stats for column q = lq (halves combined via xor-32)
+    float tm = -INFINITY;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) tm = fmaxf(tm, p[r]);
+    tm = fmaxf(tm, __shfl_xor(tm, 32, WAVE));
+    const float mnew = fmaxf(m, tm);
+    const float alpha = (mnew == -INFINITY) ? 0.f : __expf(m - mnew);
+    float ep[16];
+    float tsum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      ep[r] = (p[r] == -INFINITY) ? 0.f : __expf(p[r] - mnew);
+      tsum += ep[r];
+    }
+    tsum += __shfl_xor(tsum, 32, WAVE);
+    l = l * alpha + tsum;
+    m = mnew;
+    // pack expP to bf16 pairs and exchange halves: after this each lane can
+    // assemble A[i=q=lq][k=kv] fragments for PV
+    u32 pk[8], rcv[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      pk[i] = (u32)f2b(ep[2 * i]) | ((u32)f2b(ep[2 * i + 1]) << 16);
+      rcv[i] = __shfl_xor(pk[i], 32, WAVE);
+    }
+    // per-row alpha for the O rescale (row q via lane shuffle)
+    float arow[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      arow[r] = __shfl(alpha, (r & 3) + 8 * (r >> 2) + 4 * lhalf, WAVE);
+#pragma unroll
+    for (int db = 0; db < 4; ++db)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) oacc[db][r] *= arow[r];
+    // PV: two K=16 windows over the 32-kv tile
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      union { u32 u[4]; bf16x8 v; } af;
+      if (lhalf == 0) {
+        af.u[0] = pk[4 * kk];
+        af.u[1] = pk[4 * kk + 1];
+        af.u[2] = rcv[4 * kk];
+        af.u[3] = rcv[4 * kk + 1];
+      } else {
+        af.u[0] = rcv[4 * kk + 2];
+        af.u[1] = rcv[4 * kk + 3];
+        af.u[2] = pk[4 * kk + 2];
+        af.u[3] = pk[4 * kk + 3];
+      }
+#pragma unroll
+      for (int db = 0; db < 4; ++db) {
+        const int d = db * 32 + lq;
+        const u16* vt =
+            vtbase + (size_t)d * max_seq + pkv + 16 * kk + lhalf * 8;
+        bf16x8 vf = *reinterpret_cast<const bf16x8*>(vt);
+        oacc[db] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(af.v, vf, oacc[db], 0,
+                                                    0, 0);
+      }
+    }
+  }
+
+  // epilogue: divide by l (per q row, via shuffle) and store
+  float lrow[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    float lv = __shfl(l, (r & 3) + 8 * (r >> 2) + 4 * lhalf, WAVE);
+    lrow[r] = 1.f / lv;
+  }
+#pragma unroll
+  for (int db = 0; db < 4; ++db) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qrow = (r & 3) + 8 * (r >> 2) + 4 * lhalf;
+      const int srow = qb + qrow;
+      if (srow < S)
+        out[(size_t)srow * out_stride + (size_t)h * hd + db * 32 + lq] =
+            f2b(oacc[db][r] * lrow[r]);
+    }
   }
 }
 
@@ -1109,19 +1278,20 @@ void launch_embed_rows(const u16* embed, const u32* ids, u16* x, int S, int H,
                        hipStream_t s) {
   hipLaunchKernelGGL(k_embed_rows, dim3(S), dim3(256), 0, s, embed, ids, x, H);
 }
-void launch_rope_store_decode(u16* qkv, u16* kc, u16* vc, const float* cost,
-                              const float* sint, const int* pos, int nh,
-                              int nkv, int hd, int rd, int max_seq,
-                              hipStream_t s) {
+void launch_rope_store_decode(u16* qkv, u16* kc, u16* vc, u16* vtc,
+                              const float* cost, const float* sint,
+                              const int* pos, int nh, int nkv, int hd, int rd,
+                              int max_seq, hipStream_t s) {
   hipLaunchKernelGGL(k_rope_store_decode, dim3(nh + 2 * nkv), dim3(64), 0, s,
-                     qkv, kc, vc, cost, sint, pos, nh, nkv, hd, rd, max_seq);
+                     qkv, kc, vc, vtc, cost, sint, pos, nh, nkv, hd, rd,
+                     max_seq);
 }
-void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, const float* cost,
-                               const float* sint, int pos0, int S, int nh,
-                               int nkv, int hd, int rd, int max_seq,
-                               int qkv_stride, hipStream_t s) {
+void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, u16* vtc,
+                               const float* cost, const float* sint, int pos0,
+                               int S, int nh, int nkv, int hd, int rd,
+                               int max_seq, int qkv_stride, hipStream_t s) {
   hipLaunchKernelGGL(k_rope_store_prefill, dim3(nh + 2 * nkv, S), dim3(64), 0,
-                     s, qkv, kc, vc, cost, sint, pos0, nh, nkv, hd, rd,
+                     s, qkv, kc, vc, vtc, cost, sint, pos0, nh, nkv, hd, rd,
                      max_seq, qkv_stride);
 }
 void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
@@ -1133,12 +1303,18 @@ void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
                      nchunk);
 }
 void launch_attn_prefill(const u16* qkv, const u16* kc, const u16* vc,
-                         u16* out, int S, int pos0, int nh, int nkv, int hd,
-                         int max_seq, int qkv_stride, int out_stride,
-                         hipStream_t s) {
-  hipLaunchKernelGGL(k_attn_prefill, dim3((S + 3) / 4, nh), dim3(256), 0, s,
-                     qkv, kc, vc, out, S, pos0, nh, nkv, hd, max_seq,
-                     qkv_stride, out_stride);
+                         const u16* vtc, u16* out, int S, int pos0, int nh,
+                         int nkv, int hd, int max_seq, int qkv_stride,
+                         int out_stride, hipStream_t s) {
+  if (hd == 128) {
+    hipLaunchKernelGGL(k_attn_prefill_mfma, dim3((S + 255) / 256, nh),
+                       dim3(512), 0, s, qkv, kc, vtc, out, S, pos0, nh, nkv,
+                       max_seq, qkv_stride, out_stride);
+  } else {
+    hipLaunchKernelGGL(k_attn_prefill, dim3((S + 3) / 4, nh), dim3(256), 0, s,
+                       qkv, kc, vc, out, S, pos0, nh, nkv, hd, max_seq,
+                       qkv_stride, out_stride);
+  }
 }
 void launch_rope_simple(u16* x, const float* cost, const float* sint, int bh,
                         int s, int d, hipStream_t st) {
