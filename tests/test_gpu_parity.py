@@ -341,15 +341,15 @@ def test_max_seq_guard():
         model_type="llama", hidden_size=64, intermediate_size=128,
         vocab_size=128, num_hidden_layers=1, num_attention_heads=4,
         num_key_value_heads=2, head_dim=16, max_position_embeddings=64)
-    eng = cake_amd.Engine(json.dumps(cfg_json), max_seq=16,
+    eng = cake_amd.Engine(json.dumps(cfg_json), max_seq=32,
                           max_batch_tokens=16)
     eng.init_random(seed=1)
     try:
         with pytest.raises(cake_amd.CakeHipError, match="max_seq"):
-            eng.prefill(np.zeros(17, dtype=np.uint32))
+            eng.prefill(np.zeros(40, dtype=np.uint32))
         eng.reset()
         eng.prefill(np.zeros(8, dtype=np.uint32))
         with pytest.raises(cake_amd.CakeHipError, match="max_seq"):
-            eng.decode(20)
+            eng.decode(30)
     finally:
         eng.close()
