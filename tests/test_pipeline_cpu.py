@@ -1,0 +1,97 @@
+"""Pipeline protocol test on CPU (gloo, world_size 2): the exact rank/call
+sequence bench.py and the engine use for N>1 — rank 0: embed -> layers ->
+send -> recv -> head; rank 1: recv -> layers -> send — with the oracle as
+the per-rank compute.  Verifies the sharded pipeline reproduces the
+monolithic model's logits and greedy tokens (§8e).
+
+Uses 127.0.0.1 for rendezvous (container hostname may not resolve).
+"""
+import json
+import os
+
+import numpy as np
+import pytest
+
+from oracle import Config, OracleModel, rms_norm, linear
+from tests.helpers import fixture_weights
+
+GOLDEN = os.path.join(os.path.dirname(os.path.abspath(__file__)), "golden")
+
+
+def _rank_main(rank, world, port, result_q):
+    import torch
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    cfg_json, cfg, w, z = fixture_weights(GOLDEN, "tiny_llama3")
+    L = cfg.num_hidden_layers
+    bounds = [round(L * r / world) for r in range(world + 1)]
+    lo, hi = bounds[rank], bounds[rank + 1]
+    model = OracleModel(cfg, w)  # holds all weights; uses only [lo, hi)
+
+    prompt = list(z["prompt"])
+    gen = 6
+    tokens = list(prompt)
+    out_tokens = []
+    index_pos = 0
+    H = cfg.hidden_size
+
+    for step in range(gen):
+        ctx = tokens if step == 0 else tokens[-1:]
+        S = len(ctx)
+        if rank == 0:
+            x = w.embed_tokens[np.array([ctx])].astype(np.float32)
+            x = model.hidden_forward(x, index_pos, lo, hi)
+            dist.send(torch.from_numpy(x.copy()), dst=1)
+            x2 = torch.zeros(1, S, H)
+            dist.recv(x2, src=world - 1)
+            x = x2.numpy()
+            x = rms_norm(x, w.norm, cfg.rms_norm_eps)[:, -1, :]
+            logits = linear(x, w.lm_head)
+            nxt = int(np.argmax(logits[0]))
+            # broadcast next token so every rank appends the same id
+            t = torch.tensor([nxt], dtype=torch.int64)
+            dist.broadcast(t, src=0)
+        else:
+            x2 = torch.zeros(1, S, H)
+            dist.recv(x2, src=rank - 1)
+            x = model.hidden_forward(x2.numpy(), index_pos, lo, hi)
+            dist.send(torch.from_numpy(x.copy()), dst=(rank + 1) % world)
+            t = torch.tensor([0], dtype=torch.int64)
+            dist.broadcast(t, src=0)
+            nxt = int(t.item())
+        index_pos += S
+        tokens.append(nxt)
+        out_tokens.append(nxt)
+
+    dist.barrier()
+    dist.destroy_process_group()
+    if rank == 0:
+        result_q.put(out_tokens)
+
+
+def test_two_rank_pipeline_matches_monolithic():
+    import torch.multiprocessing as mp
+    import socket
+    # pick a free port
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_rank_main, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=180)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    cfg_json, cfg, w, z = fixture_weights(GOLDEN, "tiny_llama3")
+    ref = OracleModel(cfg, w).generate_greedy(list(z["prompt"]), 6)
+    assert got == ref, f"pipeline {got} != monolithic {ref}"
