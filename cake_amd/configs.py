@@ -35,11 +35,16 @@ QWEN3_32B = dict(
     rope_theta=1000000.0, max_position_embeddings=8192,
     tie_word_embeddings=False)
 
+QWEN3_32B_FP8 = dict(QWEN3_32B,
+    quantization_config=dict(quant_method="fp8",
+                             weight_block_size=[128, 128]))
+
 MODELS = {
     "llama3-8b": LLAMA3_8B,
     "llama3-70b": LLAMA3_70B,
     "qwen3-0.6b": QWEN3_0_6B,
     "qwen3-32b": QWEN3_32B,
+    "qwen3-32b-fp8": QWEN3_32B_FP8,
 }
 
 

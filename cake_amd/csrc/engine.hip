@@ -74,6 +74,7 @@ struct ModelConfig {
   int head_dim = 0, max_pos = 4096;
   float rms_eps = 1e-5f, rope_theta = 10000.f;
   bool tied = false, qk_norm = false;
+  bool fp8 = false;  // quantization_config.quant_method == "fp8" (fp8.rs:20-40)
   // llama3 rope scaling (config.rs:50-65)
   bool rope_llama3 = false;
   float rs_factor = 1, rs_low = 1, rs_high = 4;
@@ -111,6 +112,9 @@ static int parse_config(const char* json, ModelConfig* c) {
   if (auto p = v->get("tie_word_embeddings")) c->tied = p->bool_or(false);
   if (auto p = v->get("model_type"))
     c->qk_norm = p->str.find("qwen3") != std::string::npos;
+  if (auto qc = v->get("quantization_config"))
+    if (auto qm = qc->get("quant_method"))
+      c->fp8 = qm->str == "fp8";
   if (auto rs = v->get("rope_scaling")) {
     if (rs->kind == minijson::Value::Obj) {
       auto ty = rs->get("rope_type");
@@ -198,6 +202,10 @@ struct LayerDev {
   u16 *qnorm = nullptr, *knorm = nullptr;
   u16 *kc = nullptr, *vc = nullptr;  // (nkv, max_seq, hd) each
   u16 *vtc = nullptr;                // V transposed: (nkv, hd, max_seq)
+  // fp8 weights (e4m3fn bytes + blockwise 128x128 scale_inv, fp8.rs:42-64)
+  unsigned char *wqkv8 = nullptr, *wo8 = nullptr, *wgu8 = nullptr,
+                *wdown8 = nullptr;
+  float *sqkv = nullptr, *so8 = nullptr, *sgu = nullptr, *sdown = nullptr;
 };
 
 struct cake_engine {
@@ -216,6 +224,7 @@ struct cake_engine {
   // workspaces
   u16 *x = nullptr, *xn = nullptr, *qkv = nullptr, *attn_out = nullptr;
   u16 *gu = nullptr, *act = nullptr;
+  u16 *wscratch = nullptr;  // fp8 prefill: per-layer dequantized weight
   float* logits = nullptr;
   float* fbuf = nullptr;
   u32* ids = nullptr;
@@ -321,10 +330,14 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
   const int H = c.hidden, I = c.inter, hd = c.hd();
   const int Sq = c.sq(), Nq = c.nqkv();
   {  // rms_1 fused into the qkv projection (GEMV, x kept in registers)
-    StatScope ss(e, "gemv_qkv", (double)Nq * H * 2 + 2.0 * H * 2 + Nq * 2,
-                 2.0 * Nq * H);
-    launch_gemv(l.wqkv, e->x, e->qkv, nullptr, l.rms1, c.rms_eps, Nq, H, 0,
-                e->stream);
+    double wb = (double)Nq * H * (c.fp8 ? 1 : 2);
+    StatScope ss(e, "gemv_qkv", wb + 2.0 * H * 2 + Nq * 2, 2.0 * Nq * H);
+    if (c.fp8)
+      launch_gemv_fp8(l.wqkv8, l.sqkv, e->x, e->qkv, nullptr, l.rms1,
+                      c.rms_eps, Nq, H, 0, e->stream);
+    else
+      launch_gemv(l.wqkv, e->x, e->qkv, nullptr, l.rms1, c.rms_eps, Nq, H, 0,
+                  e->stream);
   }
   if (l.qnorm) {  // Qwen3 per-head QK-norm (attention.rs:202-215)
     StatScope ss(e, "qknorm", 2.0 * (c.nh + c.nkv) * hd * 2, 0);
@@ -347,22 +360,34 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
                        e->nchunk, e->stream);
   }
   {  // o projection + residual
-    StatScope ss(e, "gemv_o", (double)H * Sq * 2 + Sq * 2 + H * 4,
-                 2.0 * H * Sq);
-    launch_gemv(l.wo, e->attn_out, e->x, e->x, nullptr, 0.f, H, Sq, 1,
-                e->stream);
+    double wb = (double)H * Sq * (c.fp8 ? 1 : 2);
+    StatScope ss(e, "gemv_o", wb + Sq * 2 + H * 4, 2.0 * H * Sq);
+    if (c.fp8)
+      launch_gemv_fp8(l.wo8, l.so8, e->attn_out, e->x, e->x, nullptr, 0.f,
+                      H, Sq, 1, e->stream);
+    else
+      launch_gemv(l.wo, e->attn_out, e->x, e->x, nullptr, 0.f, H, Sq, 1,
+                  e->stream);
   }
   {  // rms_2 fused into gate_up GEMV + silu_mul (mlp.rs:21-31)
-    StatScope ss(e, "gemv_gateup", 2.0 * I * H * 2 + 2.0 * H * 2 + I * 2,
-                 4.0 * I * H);
-    launch_gemv_gateup(l.wgu, e->x, e->act, l.rms2, c.rms_eps, I, H,
-                       e->gu_rows, e->stream);
+    double wb = 2.0 * I * H * (c.fp8 ? 1 : 2);
+    StatScope ss(e, "gemv_gateup", wb + 2.0 * H * 2 + I * 2, 4.0 * I * H);
+    if (c.fp8)
+      launch_gemv_gateup_fp8(l.wgu8, l.sgu, e->x, e->act, l.rms2, c.rms_eps,
+                             I, H, e->stream);
+    else
+      launch_gemv_gateup(l.wgu, e->x, e->act, l.rms2, c.rms_eps, I, H,
+                         e->gu_rows, e->stream);
   }
   {  // down projection + residual
-    StatScope ss(e, "gemv_down", (double)H * I * 2 + I * 2 + H * 4,
-                 2.0 * H * I);
-    launch_gemv(l.wdown, e->act, e->x, e->x, nullptr, 0.f, H, I, 1,
-                e->stream);
+    double wb = (double)H * I * (c.fp8 ? 1 : 2);
+    StatScope ss(e, "gemv_down", wb + I * 2 + H * 4, 2.0 * H * I);
+    if (c.fp8)
+      launch_gemv_fp8(l.wdown8, l.sdown, e->act, e->x, e->x, nullptr, 0.f,
+                      H, I, 1, e->stream);
+    else
+      launch_gemv(l.wdown, e->act, e->x, e->x, nullptr, 0.f, H, I, 1,
+                  e->stream);
   }
 }
 
@@ -375,10 +400,19 @@ static void enqueue_layer_prefill(cake_engine* e, LayerDev& l, int S,
     StatScope ss(e, "rmsnorm_pf", 2.0 * S * H * 2, 0);
     launch_rmsnorm(e->x, l.rms1, e->xn, S, H, c.rms_eps, e->stream);
   }
+  const u16* wqkv = l.wqkv;
+  const u16* wo = l.wo;
+  const u16* wgu = l.wgu;
+  const u16* wdown = l.wdown;
+  if (c.fp8) {  // dequant into the shared scratch right before each GEMM
+    StatScope ss(e, "dequant_fp8", (double)Nq * H * 3, 0);
+    launch_dequant_fp8(l.wqkv8, l.sqkv, e->wscratch, Nq, H, e->stream);
+    wqkv = e->wscratch;
+  }
   {
     StatScope ss(e, "gemm_qkv", (double)Nq * H * 2 + (double)S * (H + Nq) * 2,
                  2.0 * S * Nq * H);
-    launch_gemm(e->xn, l.wqkv, e->qkv, nullptr, S, Nq, H, 0, e->stream);
+    launch_gemm(e->xn, wqkv, e->qkv, nullptr, S, Nq, H, 0, e->stream);
   }
   if (l.qnorm) {
     StatScope ss(e, "qknorm_pf", 2.0 * S * (c.nh + c.nkv) * hd * 2, 0);
@@ -400,29 +434,44 @@ static void enqueue_layer_prefill(cake_engine* e, LayerDev& l, int S,
     launch_attn_prefill(e->qkv, l.kc, l.vc, l.vtc, e->attn_out, S, pos0,
                         c.nh, c.nkv, hd, e->max_seq, Nq, Sq, e->stream);
   }
+  if (c.fp8) {
+    StatScope ss(e, "dequant_fp8", (double)H * Sq * 3, 0);
+    launch_dequant_fp8(l.wo8, l.so8, e->wscratch, H, Sq, e->stream);
+    wo = e->wscratch;
+  }
   {
     StatScope ss(e, "gemm_o", (double)H * Sq * 2 + (double)S * (Sq + H) * 2,
                  2.0 * S * H * Sq);
-    launch_gemm(e->attn_out, l.wo, e->x, e->x, S, H, Sq, 1, e->stream);
+    launch_gemm(e->attn_out, wo, e->x, e->x, S, H, Sq, 1, e->stream);
   }
   {
     StatScope ss(e, "rmsnorm_pf", 2.0 * S * H * 2, 0);
     launch_rmsnorm(e->x, l.rms2, e->xn, S, H, c.rms_eps, e->stream);
   }
+  if (c.fp8) {
+    StatScope ss(e, "dequant_fp8", 2.0 * I * H * 3, 0);
+    launch_dequant_fp8(l.wgu8, l.sgu, e->wscratch, 2 * I, H, e->stream);
+    wgu = e->wscratch;
+  }
   {
     StatScope ss(e, "gemm_gateup",
                  2.0 * I * H * 2 + (double)S * (H + 2.0 * I) * 2,
                  4.0 * S * I * H);
-    launch_gemm(e->xn, l.wgu, e->gu, nullptr, S, 2 * I, H, 0, e->stream);
+    launch_gemm(e->xn, wgu, e->gu, nullptr, S, 2 * I, H, 0, e->stream);
   }
   {
     StatScope ss(e, "silu_mul", 3.0 * S * I * 2, 0);
     launch_silu_mul_rows(e->gu, e->act, S, I, e->stream);
   }
+  if (c.fp8) {
+    StatScope ss(e, "dequant_fp8", (double)H * I * 3, 0);
+    launch_dequant_fp8(l.wdown8, l.sdown, e->wscratch, H, I, e->stream);
+    wdown = e->wscratch;
+  }
   {
     StatScope ss(e, "gemm_down", (double)H * I * 2 + (double)S * (I + H) * 2,
                  2.0 * S * H * I);
-    launch_gemm(e->act, l.wdown, e->x, e->x, S, H, I, 1, e->stream);
+    launch_gemm(e->act, wdown, e->x, e->x, S, H, I, 1, e->stream);
   }
 }
 
@@ -511,14 +560,27 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
   const int hd = c.hd(), Nq = c.nqkv(), Sq = c.sq();
   const int BT = e->bt;
   // weights
+  if (c.fp8 && (H % 128 || I % 128 || Sq % 128 || c.skv() % 128))
+    return set_err(5, "fp8 requires dims to be multiples of 128");
   e->L.resize(layer_hi - layer_lo);
   for (auto& l : e->L) {
     ALLOC(l.rms1, u16, H);
     ALLOC(l.rms2, u16, H);
-    ALLOC(l.wqkv, u16, (size_t)Nq * H);
-    ALLOC(l.wo, u16, (size_t)H * Sq);
-    ALLOC(l.wgu, u16, (size_t)2 * I * H);
-    ALLOC(l.wdown, u16, (size_t)H * I);
+    if (c.fp8) {
+      ALLOC(l.wqkv8, unsigned char, (size_t)Nq * H);
+      ALLOC(l.wo8, unsigned char, (size_t)H * Sq);
+      ALLOC(l.wgu8, unsigned char, (size_t)2 * I * H);
+      ALLOC(l.wdown8, unsigned char, (size_t)H * I);
+      ALLOC(l.sqkv, float, (size_t)(Nq / 128) * (H / 128));
+      ALLOC(l.so8, float, (size_t)(H / 128) * (Sq / 128));
+      ALLOC(l.sgu, float, (size_t)(2 * I / 128) * (H / 128));
+      ALLOC(l.sdown, float, (size_t)(H / 128) * (I / 128));
+    } else {
+      ALLOC(l.wqkv, u16, (size_t)Nq * H);
+      ALLOC(l.wo, u16, (size_t)H * Sq);
+      ALLOC(l.wgu, u16, (size_t)2 * I * H);
+      ALLOC(l.wdown, u16, (size_t)H * I);
+    }
     if (c.qk_norm) {
       ALLOC(l.qnorm, u16, hd);
       ALLOC(l.knorm, u16, hd);
@@ -553,6 +615,13 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
   ALLOC(e->attn_out, u16, (size_t)BT * Sq);
   ALLOC(e->gu, u16, (size_t)BT * 2 * I);
   ALLOC(e->act, u16, (size_t)BT * I);
+  if (c.fp8) {
+    size_t mx = std::max((size_t)Nq * H,
+                         std::max((size_t)H * Sq,
+                                  std::max((size_t)2 * I * H,
+                                           (size_t)H * I)));
+    ALLOC(e->wscratch, u16, mx);
+  }
   ALLOC(e->logits, float, V);
   ALLOC(e->fbuf, float, (size_t)BT * H);
   ALLOC(e->ids, u32, BT);
@@ -592,6 +661,10 @@ extern "C" void cake_hip_engine_free(cake_engine* e) {
     hipFree(l.rms1); hipFree(l.rms2); hipFree(l.wqkv); hipFree(l.wo);
     hipFree(l.wgu); hipFree(l.wdown); hipFree(l.kc); hipFree(l.vc);
     hipFree(l.vtc);
+    if (l.wqkv8) {
+      hipFree(l.wqkv8); hipFree(l.wo8); hipFree(l.wgu8); hipFree(l.wdown8);
+      hipFree(l.sqkv); hipFree(l.so8); hipFree(l.sgu); hipFree(l.sdown);
+    }
     if (l.qnorm) { hipFree(l.qnorm); hipFree(l.knorm); }
   }
   if (e->embed) hipFree(e->embed);
@@ -599,6 +672,7 @@ extern "C" void cake_hip_engine_free(cake_engine* e) {
   if (e->norm_w) hipFree(e->norm_w);
   hipFree(e->x); hipFree(e->xn); hipFree(e->qkv); hipFree(e->attn_out);
   hipFree(e->gu); hipFree(e->act); hipFree(e->logits); hipFree(e->fbuf);
+  if (e->wscratch) hipFree(e->wscratch);
   hipFree(e->ids); hipFree(e->ring); hipFree(e->pval); hipFree(e->pidx);
   hipFree(e->attn_ws); hipFree(e->attn_cnt);
   hipFree(e->dev_pos); hipFree(e->dev_step);
@@ -615,6 +689,44 @@ struct StTensor {
   std::vector<long> shape;
   size_t off0 = 0, off1 = 0;
 };
+
+static int upload_weight_u8(cake_engine* e, unsigned char* dst,
+                            const char* filebase, const StTensor& t,
+                            size_t expect_elems) {
+  size_t n = 1;
+  for (long d : t.shape) n *= (size_t)d;
+  if (n != expect_elems)
+    return set_err(5, "fp8 tensor shape mismatch: got %zu want %zu", n,
+                   expect_elems);
+  if (t.dtype != "F8_E4M3" && t.dtype != "U8")
+    return set_err(5, "expected F8_E4M3/U8 tensor, got %s", t.dtype.c_str());
+  HIP_TRY(hipMemcpy(dst, filebase + t.off0, n, hipMemcpyHostToDevice));
+  return 0;
+}
+
+static int upload_scale(cake_engine* e, float* dst, const char* filebase,
+                        const StTensor& t, size_t expect_elems) {
+  size_t n = 1;
+  for (long d : t.shape) n *= (size_t)d;
+  if (n != expect_elems)
+    return set_err(5, "scale_inv shape mismatch: got %zu want %zu", n,
+                   expect_elems);
+  const char* src = filebase + t.off0;
+  if (t.dtype == "F32") {
+    HIP_TRY(hipMemcpy(dst, src, n * 4, hipMemcpyHostToDevice));
+  } else if (t.dtype == "BF16") {
+    std::vector<float> tmp(n);
+    const u16* h = reinterpret_cast<const u16*>(src);
+    for (size_t i = 0; i < n; ++i) {
+      union { u32 u; float f; } v{(u32)h[i] << 16};
+      tmp[i] = v.f;
+    }
+    HIP_TRY(hipMemcpy(dst, tmp.data(), n * 4, hipMemcpyHostToDevice));
+  } else {
+    return set_err(5, "scale_inv dtype %s unsupported", t.dtype.c_str());
+  }
+  return 0;
+}
 
 static int upload_weight(cake_engine* e, u16* dst, const char* filebase,
                          const StTensor& t, size_t expect_elems) {
@@ -717,6 +829,46 @@ extern "C" int cake_hip_load_safetensors(cake_engine* e, const char* path) {
     if ((r = upload_weight(e, l.rms1, base, t, H))) return r;
     if ((r = need(pre + "post_attention_layernorm.weight", &t))) return r;
     if ((r = upload_weight(e, l.rms2, base, t, H))) return r;
+    if (c.fp8) {
+      // fp8 weights + blockwise scale_inv (fp8.rs:42-64); fused qkv and
+      // gate_up concatenate both the byte tensors and the scale rows
+      const size_t Hb = H / 128, Sqb = Sq / 128, Skvb = Skv / 128,
+                   Ib = I / 128;
+      if ((r = need(pre + "self_attn.q_proj.weight", &t))) return r;
+      if ((r = upload_weight_u8(e, l.wqkv8, base, t, Sq * H))) return r;
+      if ((r = need(pre + "self_attn.k_proj.weight", &t))) return r;
+      if ((r = upload_weight_u8(e, l.wqkv8 + Sq * H, base, t, Skv * H)))
+        return r;
+      if ((r = need(pre + "self_attn.v_proj.weight", &t))) return r;
+      if ((r = upload_weight_u8(e, l.wqkv8 + (Sq + Skv) * H, base, t,
+                                Skv * H)))
+        return r;
+      if ((r = need(pre + "self_attn.q_proj.weight_scale_inv", &t))) return r;
+      if ((r = upload_scale(e, l.sqkv, base, t, Sqb * Hb))) return r;
+      if ((r = need(pre + "self_attn.k_proj.weight_scale_inv", &t))) return r;
+      if ((r = upload_scale(e, l.sqkv + Sqb * Hb, base, t, Skvb * Hb)))
+        return r;
+      if ((r = need(pre + "self_attn.v_proj.weight_scale_inv", &t))) return r;
+      if ((r = upload_scale(e, l.sqkv + (Sqb + Skvb) * Hb, base, t,
+                            Skvb * Hb)))
+        return r;
+      if ((r = need(pre + "self_attn.o_proj.weight", &t))) return r;
+      if ((r = upload_weight_u8(e, l.wo8, base, t, H * Sq))) return r;
+      if ((r = need(pre + "self_attn.o_proj.weight_scale_inv", &t))) return r;
+      if ((r = upload_scale(e, l.so8, base, t, Hb * Sqb))) return r;
+      if ((r = need(pre + "mlp.gate_proj.weight", &t))) return r;
+      if ((r = upload_weight_u8(e, l.wgu8, base, t, I * H))) return r;
+      if ((r = need(pre + "mlp.up_proj.weight", &t))) return r;
+      if ((r = upload_weight_u8(e, l.wgu8 + I * H, base, t, I * H))) return r;
+      if ((r = need(pre + "mlp.gate_proj.weight_scale_inv", &t))) return r;
+      if ((r = upload_scale(e, l.sgu, base, t, Ib * Hb))) return r;
+      if ((r = need(pre + "mlp.up_proj.weight_scale_inv", &t))) return r;
+      if ((r = upload_scale(e, l.sgu + Ib * Hb, base, t, Ib * Hb))) return r;
+      if ((r = need(pre + "mlp.down_proj.weight", &t))) return r;
+      if ((r = upload_weight_u8(e, l.wdown8, base, t, H * I))) return r;
+      if ((r = need(pre + "mlp.down_proj.weight_scale_inv", &t))) return r;
+      if ((r = upload_scale(e, l.sdown, base, t, Hb * Ib))) return r;
+    } else {
     // fused qkv: upload q,k,v at row offsets (attention.rs:109-114)
     if ((r = need(pre + "self_attn.q_proj.weight", &t))) return r;
     if ((r = upload_weight(e, l.wqkv, base, t, Sq * H))) return r;
@@ -734,6 +886,7 @@ extern "C" int cake_hip_load_safetensors(cake_engine* e, const char* path) {
     if ((r = upload_weight(e, l.wgu + I * H, base, t, I * H))) return r;
     if ((r = need(pre + "mlp.down_proj.weight", &t))) return r;
     if ((r = upload_weight(e, l.wdown, base, t, H * I))) return r;
+    }
     if (c.qk_norm) {
       if ((r = need(pre + "self_attn.q_norm.weight", &t))) return r;
       if ((r = upload_weight(e, l.qnorm, base, t, hd))) return r;
@@ -761,13 +914,34 @@ extern "C" int cake_hip_init_random(cake_engine* e, uint64_t seed,
     launch_fill_const(e->norm_w, H, 1.0f, e->stream);
     if (!(c.tied && e->has_embed())) fill(e->lm_head, V * H);
   }
+  std::vector<float> hscales;
+  auto fill_scales = [&](float* p, size_t n) {
+    hscales.resize(n);
+    for (size_t i = 0; i < n; ++i)
+      hscales[i] = 3e-4f + 2e-4f * (float)((i * 2654435761u) % 1000) / 1000.f;
+    hipMemcpy(p, hscales.data(), n * 4, hipMemcpyHostToDevice);
+  };
   for (auto& l : e->L) {
     launch_fill_const(l.rms1, H, 1.0f, e->stream);
     launch_fill_const(l.rms2, H, 1.0f, e->stream);
-    fill(l.wqkv, (Sq + 2 * Skv) * H);
-    fill(l.wo, H * Sq);
-    fill(l.wgu, 2 * I * H);
-    fill(l.wdown, H * I);
+    if (c.fp8) {
+      const size_t Hb = H / 128, Sqb = Sq / 128, Skvb = Skv / 128,
+                   Ib = I / 128;
+      launch_fill_random_u8(l.wqkv8, (Sq + 2 * Skv) * H, seed + salt++,
+                            e->stream);
+      launch_fill_random_u8(l.wo8, H * Sq, seed + salt++, e->stream);
+      launch_fill_random_u8(l.wgu8, 2 * I * H, seed + salt++, e->stream);
+      launch_fill_random_u8(l.wdown8, H * I, seed + salt++, e->stream);
+      fill_scales(l.sqkv, (Sqb + 2 * Skvb) * Hb);
+      fill_scales(l.so8, Hb * Sqb);
+      fill_scales(l.sgu, 2 * Ib * Hb);
+      fill_scales(l.sdown, Hb * Ib);
+    } else {
+      fill(l.wqkv, (Sq + 2 * Skv) * H);
+      fill(l.wo, H * Sq);
+      fill(l.wgu, 2 * I * H);
+      fill(l.wdown, H * I);
+    }
     if (l.qnorm) {
       launch_fill_const(l.qnorm, hd, 1.0f, e->stream);
       launch_fill_const(l.knorm, hd, 1.0f, e->stream);

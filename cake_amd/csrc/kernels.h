@@ -26,6 +26,16 @@ void launch_gemv(const u16* W, const u16* x, void* out, const u16* res,
                  hipStream_t s);
 void launch_gemv_gateup(const u16* W, const u16* x, u16* out, const u16* nw,
                         float eps, int I, int K, int rows, hipStream_t s);
+void launch_gemv_fp8(const unsigned char* W, const float* sc, const u16* x,
+                     void* out, const u16* res, const u16* nw, float eps,
+                     int N, int K, int epi, hipStream_t s);
+void launch_gemv_gateup_fp8(const unsigned char* W, const float* sc,
+                            const u16* x, u16* out, const u16* nw, float eps,
+                            int I, int K, hipStream_t s);
+void launch_dequant_fp8(const unsigned char* W, const float* sc, u16* out,
+                        int N, int K, hipStream_t s);
+void launch_fill_random_u8(unsigned char* out, size_t n, uint64_t seed,
+                           hipStream_t s);
 void launch_embed_token(const u16* embed, const u32* tok, u16* x, int H,
                         hipStream_t s);
 void launch_embed_rows(const u16* embed, const u32* ids, u16* x, int S, int H,

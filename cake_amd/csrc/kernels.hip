@@ -59,6 +59,26 @@ __device__ inline short8 ntload8(const u16* p) {
   return c.s;
 }
 
+// ---------------------------------------------------------------------------
+// FP8 e4m3fn decode (OCP encoding, gfx950-native format; restates the bit
+// decode of backends/cuda/ops.cu:31-51 / utils/fp8.rs): place sign at bit
+// 31 and exp|man at bits 20-26, then scale by 2^120 — handles normals AND
+// subnormals in two ops (f32 denormals are not flushed in the default
+// kernel mode, guide §3).  The NaN encodings (0x7f/0xff) decode to a large
+// finite value; real checkpoints contain no NaN weights (documented).
+// ---------------------------------------------------------------------------
+__device__ inline float f8tof(unsigned char b) {
+  union { u32 u; float f; } v;
+  v.u = ((u32)(b & 0x80u) << 24) | ((u32)(b & 0x7fu) << 20);
+  return v.f * 0x1p120f;
+}
+using uchar16 = __attribute__((ext_vector_type(16))) unsigned char;
+__device__ inline uchar16 ntload16b(const unsigned char* p) {
+  uint4v v = __builtin_nontemporal_load(reinterpret_cast<const uint4v*>(p));
+  union { uint4v u; uchar16 c; } c{v};
+  return c.c;
+}
+
 __device__ inline float wave_sum(float v) {
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, WAVE);
@@ -99,6 +119,22 @@ __global__ void k_fill_random(u16* __restrict__ out, size_t n, uint64_t seed,
     z ^= z >> 31;
     float u = (float)(z >> 40) * (1.0f / 8388608.0f) - 1.0f;  // [-1, 1)
     out[i] = f2b(u * scale);
+  }
+}
+// random e4m3fn bytes for synthetic fp8 weights: NaN encodings (0x7f/0xff)
+// are remapped into the largest normal
+__global__ void k_fill_random_u8(unsigned char* __restrict__ out, size_t n,
+                                 uint64_t seed) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    uint64_t z = seed + 0x9e3779b97f4a7c15ull * (i + 1);
+    z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
+    z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
+    z ^= z >> 31;
+    unsigned char b = (unsigned char)(z >> 56);
+    if ((b & 0x7f) == 0x7f) b ^= 0x08;  // avoid NaN encodings
+    out[i] = b;
   }
 }
 __global__ void k_fill_const(u16* __restrict__ out, size_t n, float v) {
@@ -465,6 +501,299 @@ __global__ __launch_bounds__(256) void k_gemv_gateup(
     float g = redg[t][0] + redg[t][1] + redg[t][2] + redg[t][3];
     float u = redu[t][0] + redu[t][1] + redu[t][2] + redu[t][3];
     out[c0 + t] = f2b(g / (1.f + __expf(-g)) * u);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// FP8 GEMV family — same structure as k_gemv_reg but weights are e4m3fn
+// bytes with blockwise 128x128 scale_inv (utils/fp8.rs:42-64), dequantized
+// in-register: HALF the HBM bytes per decode step.  16 weights per 16-B
+// lane load.  KB here = ceil(K/4096) (thread covers 16 elems per iter).
+// ---------------------------------------------------------------------------
+template <int ROWS, int EPI, bool NORM, int KB>
+__global__ __launch_bounds__(256) void k_gemv_fp8(
+    const unsigned char* __restrict__ W, const float* __restrict__ sc,
+    const u16* __restrict__ x, void* __restrict__ out,
+    const u16* __restrict__ res, const u16* __restrict__ nw, float eps,
+    int N, int K, int nkb) {
+  const int t = threadIdx.x;
+  const int row0 = blockIdx.x * ROWS;
+  const int wid = t / WAVE, lane = t % WAVE;
+  __shared__ float red[ROWS > 4 ? ROWS : 4][4];
+
+  // prefetch all weight tiles (stay in flight across the x/norm phase)
+  uchar16 wpre[ROWS][KB];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r)
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 4096 + t * 16;
+      if (row0 + r < N && k0 < K)
+        wpre[r][i] = ntload16b(W + (size_t)(row0 + r) * K + k0);
+    }
+
+  // x -> registers (f32), optionally fused rms_norm
+  float xr[KB * 16];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 4096 + t * 16;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      if (k0 < K) {
+        short8 xv = *reinterpret_cast<const short8*>(x + k0 + half * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          xr[i * 16 + half * 8 + j] = b2f((u16)xv[j]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) xr[i * 16 + half * 8 + j] = 0.f;
+      }
+    }
+  }
+  if (NORM) {
+    float ss = 0.f;
+#pragma unroll
+    for (int i = 0; i < KB * 16; ++i) ss += xr[i] * xr[i];
+    ss = wave_sum(ss);
+    if (lane == 0) red[0][wid] = ss;
+    __syncthreads();
+    const float scale =
+        rsqrtf((red[0][0] + red[0][1] + red[0][2] + red[0][3]) / (float)K +
+               eps);
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 4096 + t * 16;
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        if (k0 < K) {
+          short8 wv =
+              *reinterpret_cast<const short8*>(nw + k0 + half * 8);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int ix = i * 16 + half * 8 + j;
+            xr[ix] = b2f(f2b(xr[ix] * scale * b2f((u16)wv[j])));
+          }
+        }
+      }
+    }
+  }
+
+  float acc[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) acc[r] = 0.f;
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r)
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 4096 + t * 16;
+      if (row0 + r < N && k0 < K) {
+        // one scale_inv block covers this thread's 16 k's (k0 % 16 == 0)
+        const float s =
+            sc[(size_t)((row0 + r) / 128) * nkb + (k0 / 128)];
+        float a = 0.f;
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          a = fmaf(f8tof(wpre[r][i][j]), xr[i * 16 + j], a);
+        acc[r] = fmaf(a, s, acc[r]);
+      }
+    }
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) {
+    float v = wave_sum(acc[r]);
+    if (lane == 0) red[r][wid] = v;
+  }
+  __syncthreads();
+  if (t < ROWS) {
+    const int row = row0 + t;
+    if (row < N) {
+      float v = red[t][0] + red[t][1] + red[t][2] + red[t][3];
+      if (EPI == 2) {
+        reinterpret_cast<float*>(out)[row] = v;
+      } else if (EPI == 1) {
+        reinterpret_cast<u16*>(out)[row] = f2b(v + b2f(res[row]));
+      } else {
+        reinterpret_cast<u16*>(out)[row] = f2b(v);
+      }
+    }
+  }
+}
+
+// fp8 streaming fallback for K > 16384 (e.g. Qwen3-32B down K=25600)
+template <int ROWS, int EPI>
+__global__ __launch_bounds__(256) void k_gemv_fp8_stream(
+    const unsigned char* __restrict__ W, const float* __restrict__ sc,
+    const u16* __restrict__ x, void* __restrict__ out,
+    const u16* __restrict__ res, int N, int K, int nkb) {
+  const int t = threadIdx.x;
+  const int row0 = blockIdx.x * ROWS;
+  const int wid = t / WAVE, lane = t % WAVE;
+  float acc[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) acc[r] = 0.f;
+  const int kiter = (K + 4095) / 4096;
+#pragma unroll 1
+  for (int i = 0; i < kiter; ++i) {
+    const int k0 = i * 4096 + t * 16;
+    if (k0 >= K) continue;
+    float xv[16];
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      short8 xs = *reinterpret_cast<const short8*>(x + k0 + half * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xv[half * 8 + j] = b2f((u16)xs[j]);
+    }
+    uchar16 wv[ROWS];
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+      if (row0 + r < N) wv[r] = ntload16b(W + (size_t)(row0 + r) * K + k0);
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+      if (row0 + r < N) {
+        const float s = sc[(size_t)((row0 + r) / 128) * nkb + (k0 / 128)];
+        float a = 0.f;
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          a = fmaf(f8tof(wv[r][j]), xv[j], a);
+        acc[r] = fmaf(a, s, acc[r]);
+      }
+  }
+  __shared__ float red[ROWS > 4 ? ROWS : 4][4];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) {
+    float v = wave_sum(acc[r]);
+    if (lane == 0) red[r][wid] = v;
+  }
+  __syncthreads();
+  if (t < ROWS) {
+    const int row = row0 + t;
+    if (row < N) {
+      float v = red[t][0] + red[t][1] + red[t][2] + red[t][3];
+      if (EPI == 2) {
+        reinterpret_cast<float*>(out)[row] = v;
+      } else if (EPI == 1) {
+        reinterpret_cast<u16*>(out)[row] = f2b(v + b2f(res[row]));
+      } else {
+        reinterpret_cast<u16*>(out)[row] = f2b(v);
+      }
+    }
+  }
+}
+
+// fp8 gate_up + silu_mul (the fused MLP front half, fp8 weights)
+template <int ROWS, bool NORM, int KB>
+__global__ __launch_bounds__(256) void k_gemv_gateup_fp8(
+    const unsigned char* __restrict__ W, const float* __restrict__ sc,
+    const u16* __restrict__ x, u16* __restrict__ out,
+    const u16* __restrict__ nw, float eps, int I, int K, int nkb) {
+  const int t = threadIdx.x;
+  const int c0 = blockIdx.x * ROWS;
+  const int wid = t / WAVE, lane = t % WAVE;
+  __shared__ float redg[8][4], redu[8][4];
+
+  float xr[KB * 16];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 4096 + t * 16;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      if (k0 < K) {
+        short8 xv = *reinterpret_cast<const short8*>(x + k0 + half * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          xr[i * 16 + half * 8 + j] = b2f((u16)xv[j]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) xr[i * 16 + half * 8 + j] = 0.f;
+      }
+    }
+  }
+  if (NORM) {
+    float ss = 0.f;
+#pragma unroll
+    for (int i = 0; i < KB * 16; ++i) ss += xr[i] * xr[i];
+    ss = wave_sum(ss);
+    if (lane == 0) redg[0][wid] = ss;
+    __syncthreads();
+    const float scale =
+        rsqrtf((redg[0][0] + redg[0][1] + redg[0][2] + redg[0][3]) /
+                   (float)K + eps);
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 4096 + t * 16;
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        if (k0 < K) {
+          short8 wv =
+              *reinterpret_cast<const short8*>(nw + k0 + half * 8);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int ix = i * 16 + half * 8 + j;
+            xr[ix] = b2f(f2b(xr[ix] * scale * b2f((u16)wv[j])));
+          }
+        }
+      }
+    }
+  }
+
+  float accg[ROWS], accu[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) accg[r] = accu[r] = 0.f;
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 4096 + t * 16;
+    if (k0 >= K) continue;
+    uchar16 gv[ROWS], uv[ROWS];
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+      if (c0 + r < I) {
+        gv[r] = ntload16b(W + (size_t)(c0 + r) * K + k0);
+        uv[r] = ntload16b(W + (size_t)(c0 + r + I) * K + k0);
+      }
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+      if (c0 + r < I) {
+        const float sg = sc[(size_t)((c0 + r) / 128) * nkb + (k0 / 128)];
+        const float su =
+            sc[(size_t)((c0 + r + I) / 128) * nkb + (k0 / 128)];
+        float g = 0.f, u = 0.f;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          g = fmaf(f8tof(gv[r][j]), xr[i * 16 + j], g);
+          u = fmaf(f8tof(uv[r][j]), xr[i * 16 + j], u);
+        }
+        accg[r] = fmaf(g, sg, accg[r]);
+        accu[r] = fmaf(u, su, accu[r]);
+      }
+  }
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) {
+    float g = wave_sum(accg[r]);
+    float u = wave_sum(accu[r]);
+    if (lane == 0) { redg[r][wid] = g; redu[r][wid] = u; }
+  }
+  __syncthreads();
+  if (t < ROWS && c0 + t < I) {
+    float g = redg[t][0] + redg[t][1] + redg[t][2] + redg[t][3];
+    float u = redu[t][0] + redu[t][1] + redu[t][2] + redu[t][3];
+    out[c0 + t] = f2b(g / (1.f + __expf(-g)) * u);
+  }
+}
+
+// fp8 -> bf16 blockwise dequant (prefill path: dequant the layer's weight
+// into a scratch buffer, then run the bf16 MFMA GEMM — cake itself
+// dequantizes at load time, fp8.rs:42-64; we dequant per layer per prefill
+// to keep HBM fp8-resident for decode)
+__global__ void k_dequant_fp8(const unsigned char* __restrict__ W,
+                              const float* __restrict__ sc,
+                              u16* __restrict__ out, int N, int K, int nkb) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  size_t total = (size_t)N * K;
+  for (; i < total; i += stride) {
+    int r = (int)(i / K), k = (int)(i % K);
+    out[i] = f2b(f8tof(W[i]) * sc[(size_t)(r / 128) * nkb + k / 128]);
   }
 }
 
@@ -1167,6 +1496,11 @@ void launch_fill_random(u16* out, size_t n, uint64_t seed, float scale,
   hipLaunchKernelGGL(k_fill_random, dim3(2048), dim3(256), 0, s, out, n, seed,
                      scale);
 }
+void launch_fill_random_u8(unsigned char* out, size_t n, uint64_t seed,
+                           hipStream_t s) {
+  hipLaunchKernelGGL(k_fill_random_u8, dim3(2048), dim3(256), 0, s, out, n,
+                     seed);
+}
 void launch_fill_const(u16* out, size_t n, float v, hipStream_t s) {
   hipLaunchKernelGGL(k_fill_const, dim3(512), dim3(256), 0, s, out, n, v);
 }
@@ -1269,6 +1603,63 @@ void launch_gemv_gateup(const u16* W, const u16* x, u16* out, const u16* nw,
   else GU_KB(8);
 #undef GU_KB
 #undef GU_KB2
+}
+void launch_gemv_fp8(const unsigned char* W, const float* sc, const u16* x,
+                     void* out, const u16* res, const u16* nw, float eps,
+                     int N, int K, int epi, hipStream_t s) {
+  const int nkb = (K + 127) / 128;
+  const int rows = N >= 16384 ? 8 : 4;
+#define F8_KB(R, EPI, KB)                                                   \
+  do {                                                                      \
+    dim3 grid((N + R - 1) / R);                                             \
+    if (nw)                                                                 \
+      hipLaunchKernelGGL((k_gemv_fp8<R, EPI, true, KB>), grid, dim3(256),   \
+                         0, s, W, sc, x, out, res, nw, eps, N, K, nkb);     \
+    else                                                                    \
+      hipLaunchKernelGGL((k_gemv_fp8<R, EPI, false, KB>), grid, dim3(256),  \
+                         0, s, W, sc, x, out, res, nw, eps, N, K, nkb);     \
+  } while (0)
+#define F8_R(R, EPI)                                                        \
+  do {                                                                      \
+    if (K > 16384) {                                                        \
+      dim3 grid((N + R - 1) / R);                                           \
+      hipLaunchKernelGGL((k_gemv_fp8_stream<R, EPI>), grid, dim3(256), 0,   \
+                         s, W, sc, x, out, res, N, K, nkb);                 \
+    } else if (K <= 4096) F8_KB(R, EPI, 1);                                 \
+    else if (K <= 8192) F8_KB(R, EPI, 2);                                   \
+    else F8_KB(R, EPI, 4);                                                  \
+  } while (0)
+  if (epi == 0) { if (rows == 8) F8_R(8, 0); else F8_R(4, 0); }
+  else if (epi == 1) { if (rows == 8) F8_R(8, 1); else F8_R(4, 1); }
+  else { if (rows == 8) F8_R(8, 2); else F8_R(4, 2); }
+#undef F8_R
+#undef F8_KB
+}
+void launch_gemv_gateup_fp8(const unsigned char* W, const float* sc,
+                            const u16* x, u16* out, const u16* nw, float eps,
+                            int I, int K, hipStream_t s) {
+  const int nkb = (K + 127) / 128;
+  dim3 grid((I + 3) / 4);
+#define GU8_KB(KB)                                                          \
+  do {                                                                      \
+    if (nw)                                                                 \
+      hipLaunchKernelGGL((k_gemv_gateup_fp8<4, true, KB>), grid, dim3(256), \
+                         0, s, W, sc, x, out, nw, eps, I, K, nkb);          \
+    else                                                                    \
+      hipLaunchKernelGGL((k_gemv_gateup_fp8<4, false, KB>), grid,           \
+                         dim3(256), 0, s, W, sc, x, out, nw, eps, I, K,     \
+                         nkb);                                              \
+  } while (0)
+  if (K <= 4096) GU8_KB(1);
+  else if (K <= 8192) GU8_KB(2);
+  else GU8_KB(4);
+#undef GU8_KB
+}
+void launch_dequant_fp8(const unsigned char* W, const float* sc, u16* out,
+                        int N, int K, hipStream_t s) {
+  const int nkb = (K + 127) / 128;
+  hipLaunchKernelGGL(k_dequant_fp8, dim3(4096), dim3(256), 0, s, W, sc, out,
+                     N, K, nkb);
 }
 void launch_embed_token(const u16* embed, const u32* tok, u16* x, int H,
                         hipStream_t s) {

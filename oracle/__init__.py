@@ -389,3 +389,40 @@ def random_weights(cfg: Config, seed: int = 299792458,
         lm_head=embed if cfg.tie_word_embeddings else t(V, H),
         layers=layers,
     )
+
+
+# ---------------------------------------------------------------------------
+# FP8 (float8_e4m3fn) blockwise dequantization — restates utils/fp8.rs:42-64:
+#   w[i, j] = f8_to_f32(w8[i, j]) * scale_inv[i // 128, j // 128]
+# E4M3FN: 1 sign, 4 exp (bias 7), 3 mantissa; NaN = 0x7F/0xFF; no inf.
+# (gfx950 uses the same OCP e4m3fn encoding.)
+# ---------------------------------------------------------------------------
+FP8_BLOCK = 128  # fp8.rs:17
+
+
+def f8e4m3_to_f32(b: np.ndarray) -> np.ndarray:
+    """Decode uint8 e4m3fn -> f32 (vectorized restatement of the bit
+    decode in backends/cuda/ops.cu:31-51)."""
+    b = b.astype(np.uint32)
+    sign = np.where(b & 0x80, np.float32(-1.0), np.float32(1.0))
+    e = (b >> 3) & 0xF
+    m = b & 0x7
+    nan = (e == 15) & (m == 7)
+    val = np.where(
+        e == 0,
+        m.astype(np.float32) * np.float32(2.0 ** -9),          # subnormal
+        (1.0 + m.astype(np.float32) / 8.0) *
+        np.exp2((e.astype(np.float32) - 7.0)))
+    out = (sign * val).astype(np.float32)
+    out[nan] = np.nan
+    return out
+
+
+def fp8_dequant(w8: np.ndarray, scale_inv: np.ndarray) -> np.ndarray:
+    """utils/fp8.rs:42-64 — blockwise 128x128 dequant to f32."""
+    n, k = w8.shape
+    f = f8e4m3_to_f32(w8)
+    bi = np.arange(n) // FP8_BLOCK
+    bj = np.arange(k) // FP8_BLOCK
+    return (f * scale_inv.astype(np.float32)[np.ix_(bi, bj)]).astype(
+        np.float32)

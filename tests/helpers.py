@@ -76,3 +76,64 @@ def quantize_bf16(a: np.ndarray) -> np.ndarray:
     nan = (u & 0x7FFFFFFF) > 0x7F800000
     r = np.where(nan, np.uint32(0x7FC00000), r & 0xFFFF0000)
     return r.astype(np.uint32).view(np.float32).reshape(a.shape)
+
+
+def random_fp8_model(cfg: Config, seed: int = 7):
+    """Synthetic fp8 (e4m3fn + blockwise scale_inv) model: returns
+    (tensors dict for safetensors, dequantized-f32 ModelWeights for the
+    oracle).  Restates utils/fp8.rs:42-64 blockwise semantics."""
+    from oracle import fp8_dequant
+    rng = np.random.default_rng(seed)
+    H, I, V = cfg.hidden_size, cfg.intermediate_size, cfg.vocab_size
+    hd, nh, nkv = cfg.hd, cfg.num_attention_heads, cfg.num_key_value_heads
+
+    def w8(n, k):
+        b = rng.integers(0, 256, size=(n, k)).astype(np.uint8)
+        b[(b & 0x7F) == 0x7F] ^= 0x08  # no NaN encodings
+        sc = (3e-4 + 2e-4 * rng.random(((n + 127) // 128, (k + 127) // 128))
+              ).astype(np.float32)
+        return b, sc
+
+    def f32(*shape):
+        return (rng.standard_normal(shape) * 0.02).astype(np.float32)
+
+    tensors = {}
+    layers = []
+    for i in range(cfg.num_hidden_layers):
+        p = f"model.layers.{i}."
+        parts = {}
+        for name, n, k in (("self_attn.q_proj", nh * hd, H),
+                           ("self_attn.k_proj", nkv * hd, H),
+                           ("self_attn.v_proj", nkv * hd, H),
+                           ("self_attn.o_proj", H, nh * hd),
+                           ("mlp.gate_proj", I, H),
+                           ("mlp.up_proj", I, H),
+                           ("mlp.down_proj", H, I)):
+            b, sc = w8(n, k)
+            tensors[p + name + ".weight"] = b
+            tensors[p + name + ".weight_scale_inv"] = sc
+            parts[name] = fp8_dequant(b, sc)
+        ln1, ln2 = 1.0 + f32(H), 1.0 + f32(H)
+        tensors[p + "input_layernorm.weight"] = ln1
+        tensors[p + "post_attention_layernorm.weight"] = ln2
+        layers.append(LayerWeights(
+            input_layernorm=ln1, post_attention_layernorm=ln2,
+            q_proj=parts["self_attn.q_proj"], k_proj=parts["self_attn.k_proj"],
+            v_proj=parts["self_attn.v_proj"], o_proj=parts["self_attn.o_proj"],
+            gate_proj=parts["mlp.gate_proj"], up_proj=parts["mlp.up_proj"],
+            down_proj=parts["mlp.down_proj"]))
+    embed = f32(V, H)
+    norm = 1.0 + f32(H)
+    head = embed if cfg.tie_word_embeddings else f32(V, H)
+    tensors["model.embed_tokens.weight"] = embed
+    tensors["model.norm.weight"] = norm
+    if not cfg.tie_word_embeddings:
+        tensors["lm_head.weight"] = head
+    w = ModelWeights(embed_tokens=embed, norm=norm, lm_head=head,
+                     layers=layers)
+    return tensors, w
+
+
+def save_safetensors_raw(tensors: dict, path: str):
+    from safetensors.numpy import save_file
+    save_file({k: np.ascontiguousarray(v) for k, v in tensors.items()}, path)
