@@ -72,11 +72,22 @@ __device__ inline float f8tof(unsigned char b) {
   v.u = ((u32)(b & 0x80u) << 24) | ((u32)(b & 0x7fu) << 20);
   return v.f * 0x1p120f;
 }
-using uchar16 = __attribute__((ext_vector_type(16))) unsigned char;
-__device__ inline uchar16 ntload16b(const unsigned char* p) {
-  uint4v v = __builtin_nontemporal_load(reinterpret_cast<const uint4v*>(p));
-  union { uint4v u; uchar16 c; } c{v};
-  return c.c;
+using f32x2 = __attribute__((ext_vector_type(2))) float;
+__device__ inline uint4v ntload16b(const unsigned char* p) {
+  return __builtin_nontemporal_load(reinterpret_cast<const uint4v*>(p));
+}
+// 16 fp8 -> 16 f32 via the hardware v_cvt_pk_f32_fp8 (2 elems/instruction;
+// a bit-twiddle decode is VALU-bound at ~1 TB/s)
+__device__ inline void f8x16_decode(uint4v w, float* out) {
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(w[q], false);
+    f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(w[q], true);
+    out[q * 4 + 0] = lo[0];
+    out[q * 4 + 1] = lo[1];
+    out[q * 4 + 2] = hi[0];
+    out[q * 4 + 3] = hi[1];
+  }
 }
 
 __device__ inline float wave_sum(float v) {
@@ -522,7 +533,7 @@ __global__ __launch_bounds__(256) void k_gemv_fp8(
   __shared__ float red[ROWS > 4 ? ROWS : 4][4];
 
   // prefetch all weight tiles (stay in flight across the x/norm phase)
-  uchar16 wpre[ROWS][KB];
+  uint4v wpre[ROWS][KB];
 #pragma unroll
   for (int r = 0; r < ROWS; ++r)
 #pragma unroll
@@ -591,10 +602,12 @@ __global__ __launch_bounds__(256) void k_gemv_fp8(
         // one scale_inv block covers this thread's 16 k's (k0 % 16 == 0)
         const float s =
             sc[(size_t)((row0 + r) / 128) * nkb + (k0 / 128)];
+        float wd[16];
+        f8x16_decode(wpre[r][i], wd);
         float a = 0.f;
 #pragma unroll
         for (int j = 0; j < 16; ++j)
-          a = fmaf(f8tof(wpre[r][i][j]), xr[i * 16 + j], a);
+          a = fmaf(wd[j], xr[i * 16 + j], a);
         acc[r] = fmaf(a, s, acc[r]);
       }
     }
@@ -643,7 +656,7 @@ __global__ __launch_bounds__(256) void k_gemv_fp8_stream(
 #pragma unroll
       for (int j = 0; j < 8; ++j) xv[half * 8 + j] = b2f((u16)xs[j]);
     }
-    uchar16 wv[ROWS];
+    uint4v wv[ROWS];
 #pragma unroll
     for (int r = 0; r < ROWS; ++r)
       if (row0 + r < N) wv[r] = ntload16b(W + (size_t)(row0 + r) * K + k0);
@@ -651,10 +664,12 @@ __global__ __launch_bounds__(256) void k_gemv_fp8_stream(
     for (int r = 0; r < ROWS; ++r)
       if (row0 + r < N) {
         const float s = sc[(size_t)((row0 + r) / 128) * nkb + (k0 / 128)];
+        float wd[16];
+        f8x16_decode(wv[r], wd);
         float a = 0.f;
 #pragma unroll
         for (int j = 0; j < 16; ++j)
-          a = fmaf(f8tof(wv[r][j]), xv[j], a);
+          a = fmaf(wd[j], xv[j], a);
         acc[r] = fmaf(a, s, acc[r]);
       }
   }
@@ -744,7 +759,7 @@ __global__ __launch_bounds__(256) void k_gemv_gateup_fp8(
   for (int i = 0; i < KB; ++i) {
     const int k0 = i * 4096 + t * 16;
     if (k0 >= K) continue;
-    uchar16 gv[ROWS], uv[ROWS];
+    uint4v gv[ROWS], uv[ROWS];
 #pragma unroll
     for (int r = 0; r < ROWS; ++r)
       if (c0 + r < I) {
@@ -757,11 +772,14 @@ __global__ __launch_bounds__(256) void k_gemv_gateup_fp8(
         const float sg = sc[(size_t)((c0 + r) / 128) * nkb + (k0 / 128)];
         const float su =
             sc[(size_t)((c0 + r + I) / 128) * nkb + (k0 / 128)];
+        float gd[16], ud[16];
+        f8x16_decode(gv[r], gd);
+        f8x16_decode(uv[r], ud);
         float g = 0.f, u = 0.f;
 #pragma unroll
         for (int j = 0; j < 16; ++j) {
-          g = fmaf(f8tof(gv[r][j]), xr[i * 16 + j], g);
-          u = fmaf(f8tof(uv[r][j]), xr[i * 16 + j], u);
+          g = fmaf(gd[j], xr[i * 16 + j], g);
+          u = fmaf(ud[j], xr[i * 16 + j], u);
         }
         accg[r] = fmaf(g, sg, accg[r]);
         accu[r] = fmaf(u, su, accu[r]);
@@ -793,7 +811,8 @@ __global__ void k_dequant_fp8(const unsigned char* __restrict__ W,
   size_t total = (size_t)N * K;
   for (; i < total; i += stride) {
     int r = (int)(i / K), k = (int)(i % K);
-    out[i] = f2b(f8tof(W[i]) * sc[(size_t)(r / 128) * nkb + k / 128]);
+    f32x2 v = __builtin_amdgcn_cvt_pk_f32_fp8((u32)W[i], false);
+    out[i] = f2b(v[0] * sc[(size_t)(r / 128) * nkb + k / 128]);
   }
 }
 
