@@ -339,17 +339,11 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
       launch_gemv(l.wqkv, e->x, e->qkv, nullptr, l.rms1, c.rms_eps, Nq, H, 0,
                   e->stream);
   }
-  if (l.qnorm) {  // Qwen3 per-head QK-norm (attention.rs:202-215)
-    StatScope ss(e, "qknorm", 2.0 * (c.nh + c.nkv) * hd * 2, 0);
-    launch_rmsnorm(e->qkv, l.qnorm, e->qkv, c.nh, hd, c.rms_eps, e->stream);
-    launch_rmsnorm(e->qkv + Sq, l.knorm, e->qkv + Sq, c.nkv, hd, c.rms_eps,
-                   e->stream);
-  }
-  {  // rope q,k + KV store at slot *pos
+  {  // [Qwen3 QK-norm, attention.rs:202-215, fused +] rope + KV store
     StatScope ss(e, "rope_store", (double)Nq * hd * 0, 0);
     launch_rope_store_decode(e->qkv, l.kc, l.vc, l.vtc, e->cos_t, e->sin_t,
                              e->dev_pos, c.nh, c.nkv, hd, hd, e->max_seq,
-                             e->stream);
+                             l.qnorm, l.knorm, c.rms_eps, e->stream);
   }
   {  // decode attention over the cache (single launch, split-KV combine)
     double kvbytes = 2.0 * (e->host_pos + 1) * c.skv() * 2;
@@ -414,18 +408,11 @@ static void enqueue_layer_prefill(cake_engine* e, LayerDev& l, int S,
                  2.0 * S * Nq * H);
     launch_gemm(e->xn, wqkv, e->qkv, nullptr, S, Nq, H, 0, e->stream);
   }
-  if (l.qnorm) {
-    StatScope ss(e, "qknorm_pf", 2.0 * S * (c.nh + c.nkv) * hd * 2, 0);
-    launch_rmsnorm_strided(e->qkv, l.qnorm, e->qkv, S, c.nh, (size_t)Nq, hd,
-                           c.rms_eps, e->stream);
-    launch_rmsnorm_strided(e->qkv + Sq, l.knorm, e->qkv + Sq, S, c.nkv,
-                           (size_t)Nq, hd, c.rms_eps, e->stream);
-  }
-  {
+  {  // [fused QK-norm +] rope + KV store for all S positions
     StatScope ss(e, "rope_store_pf", 0, 0);
     launch_rope_store_prefill(e->qkv, l.kc, l.vc, l.vtc, e->cos_t, e->sin_t,
                               pos0, S, c.nh, c.nkv, hd, hd, e->max_seq, Nq,
-                              e->stream);
+                              l.qnorm, l.knorm, c.rms_eps, e->stream);
   }
   {
     double n_avg = pos0 + (S + 1) * 0.5;

@@ -845,31 +845,56 @@ __global__ void k_embed_rows(const u16* __restrict__ embed,
 // *pos (replaces cache.rs:195-196 cat).  qkv layout: (Sq | Skv | Skv) per row.
 // grid = nh + 2*nkv blocks.
 // ---------------------------------------------------------------------------
+// Optional fused per-head QK-norm (attention.rs:202-215): applied to the
+// head row before the rotation, re-quantized to bf16 so it matches the
+// separate rmsnorm-kernel path bit-exactly.  One wave per head.
+__device__ inline float head_norm_scale(const u16* row, int hd, float eps) {
+  float ss = 0.f;
+  for (int i = threadIdx.x; i < hd; i += 64) {
+    float f = b2f(row[i]);
+    ss += f * f;
+  }
+  ss = wave_sum(ss);
+  ss = __shfl(ss, 0, WAVE);
+  return rsqrtf(ss / (float)hd + eps);
+}
+__device__ inline float nrm(const u16* row, const u16* w, int i, float sc,
+                            bool on) {
+  float v = b2f(row[i]);
+  return on ? b2f(f2b(v * sc * b2f(w[i]))) : v;
+}
+
 __global__ void k_rope_store_decode(u16* __restrict__ qkv,
                                     u16* __restrict__ kc, u16* __restrict__ vc,
                                     u16* __restrict__ vtc,
                                     const float* __restrict__ cost,
                                     const float* __restrict__ sint,
                                     const int* __restrict__ pos, int nh,
-                                    int nkv, int hd, int rd, int max_seq) {
+                                    int nkv, int hd, int rd, int max_seq,
+                                    const u16* __restrict__ qn,
+                                    const u16* __restrict__ kn, float eps) {
   const int b = blockIdx.x;
   const int p = *pos;
   const int half = rd / 2;
   const float* c = cost + (size_t)p * half;
   const float* s = sint + (size_t)p * half;
-  if (b < nh) {                       // rope q head in place
+  if (b < nh) {                       // [qk-norm +] rope q head in place
     u16* q = qkv + (size_t)b * hd;
+    const bool on = qn != nullptr;
+    const float sc = on ? head_norm_scale(q, hd, eps) : 1.f;
     for (int i = threadIdx.x; i < half; i += blockDim.x) {
-      float x1 = b2f(q[i]), x2 = b2f(q[i + half]);
+      float x1 = nrm(q, qn, i, sc, on), x2 = nrm(q, qn, i + half, sc, on);
       q[i] = f2b(x1 * c[i] - x2 * s[i]);
       q[i + half] = f2b(x2 * c[i] + x1 * s[i]);
     }
-  } else if (b < nh + nkv) {          // rope k head -> cache slot p
+  } else if (b < nh + nkv) {          // [qk-norm +] rope k head -> slot p
     const int h = b - nh;
     u16* k = qkv + (size_t)(nh + h) * hd;
     u16* dst = kc + ((size_t)h * max_seq + p) * hd;
+    const bool on = kn != nullptr;
+    const float sc = on ? head_norm_scale(k, hd, eps) : 1.f;
     for (int i = threadIdx.x; i < half; i += blockDim.x) {
-      float x1 = b2f(k[i]), x2 = b2f(k[i + half]);
+      float x1 = nrm(k, kn, i, sc, on), x2 = nrm(k, kn, i + half, sc, on);
       dst[i] = f2b(x1 * c[i] - x2 * s[i]);
       dst[i + half] = f2b(x2 * c[i] + x1 * s[i]);
     }
@@ -894,7 +919,9 @@ __global__ void k_rope_store_prefill(u16* __restrict__ qkv,
                                      const float* __restrict__ cost,
                                      const float* __restrict__ sint, int pos0,
                                      int nh, int nkv, int hd, int rd,
-                                     int max_seq, int qkv_stride) {
+                                     int max_seq, int qkv_stride,
+                                     const u16* __restrict__ qn,
+                                     const u16* __restrict__ kn, float eps) {
   const int b = blockIdx.x;
   const int sidx = blockIdx.y;
   const int p = pos0 + sidx;
@@ -904,8 +931,10 @@ __global__ void k_rope_store_prefill(u16* __restrict__ qkv,
   u16* row = qkv + (size_t)sidx * qkv_stride;
   if (b < nh) {
     u16* q = row + (size_t)b * hd;
+    const bool on = qn != nullptr;
+    const float sc = on ? head_norm_scale(q, hd, eps) : 1.f;
     for (int i = threadIdx.x; i < half; i += blockDim.x) {
-      float x1 = b2f(q[i]), x2 = b2f(q[i + half]);
+      float x1 = nrm(q, qn, i, sc, on), x2 = nrm(q, qn, i + half, sc, on);
       q[i] = f2b(x1 * c[i] - x2 * s[i]);
       q[i + half] = f2b(x2 * c[i] + x1 * s[i]);
     }
@@ -913,8 +942,10 @@ __global__ void k_rope_store_prefill(u16* __restrict__ qkv,
     const int h = b - nh;
     u16* k = row + (size_t)(nh + h) * hd;
     u16* dst = kc + ((size_t)h * max_seq + p) * hd;
+    const bool on = kn != nullptr;
+    const float sc = on ? head_norm_scale(k, hd, eps) : 1.f;
     for (int i = threadIdx.x; i < half; i += blockDim.x) {
-      float x1 = b2f(k[i]), x2 = b2f(k[i + half]);
+      float x1 = nrm(k, kn, i, sc, on), x2 = nrm(k, kn, i + half, sc, on);
       dst[i] = f2b(x1 * c[i] - x2 * s[i]);
       dst[i + half] = f2b(x2 * c[i] + x1 * s[i]);
     }
@@ -1691,18 +1722,20 @@ void launch_embed_rows(const u16* embed, const u32* ids, u16* x, int S, int H,
 void launch_rope_store_decode(u16* qkv, u16* kc, u16* vc, u16* vtc,
                               const float* cost, const float* sint,
                               const int* pos, int nh, int nkv, int hd, int rd,
-                              int max_seq, hipStream_t s) {
+                              int max_seq, const u16* qn, const u16* kn,
+                              float eps, hipStream_t s) {
   hipLaunchKernelGGL(k_rope_store_decode, dim3(nh + 2 * nkv), dim3(64), 0, s,
                      qkv, kc, vc, vtc, cost, sint, pos, nh, nkv, hd, rd,
-                     max_seq);
+                     max_seq, qn, kn, eps);
 }
 void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, u16* vtc,
                                const float* cost, const float* sint, int pos0,
                                int S, int nh, int nkv, int hd, int rd,
-                               int max_seq, int qkv_stride, hipStream_t s) {
+                               int max_seq, int qkv_stride, const u16* qn,
+                               const u16* kn, float eps, hipStream_t s) {
   hipLaunchKernelGGL(k_rope_store_prefill, dim3(nh + 2 * nkv, S), dim3(64), 0,
                      s, qkv, kc, vc, vtc, cost, sint, pos0, nh, nkv, hd, rd,
-                     max_seq, qkv_stride);
+                     max_seq, qkv_stride, qn, kn, eps);
 }
 void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
                         const int* pos, float* ws, u32* cnt, u16* out, int nh,
