@@ -1432,8 +1432,21 @@ __global__ __launch_bounds__(256) void k_gemm_bf16(
   u16* As = reinterpret_cast<u16*>(smem);                       // [2][128][64]
   u16* Bs = reinterpret_cast<u16*>(smem + 2 * GEMM_BM * GEMM_BK * 2);
 
-  const int m0 = blockIdx.x * GEMM_BM;
-  const int n0 = blockIdx.y * GEMM_BN;
+  // XCD-aware block swizzle (guide T1, bijective variant): contiguous
+  // chunks of the grid per XCD so neighbor tiles hit the same per-XCD L2
+  int mt, nt;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    const int orig = blockIdx.x + gridDim.x * blockIdx.y;
+    const int q = nwg / 8, rr = nwg % 8;
+    const int xcd = orig % 8, idx = orig / 8;
+    const int wgid =
+        (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+    mt = wgid % gridDim.x;
+    nt = wgid / gridDim.x;
+  }
+  const int m0 = mt * GEMM_BM;
+  const int n0 = nt * GEMM_BN;
   const int t = threadIdx.x;
   const int wid = t / WAVE, lane = t % WAVE;
   const int wr = wid / 2, wc = wid % 2;  // wave's 64x64 quadrant
