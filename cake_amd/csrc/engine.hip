@@ -575,6 +575,11 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
     ALLOC(l.kc, u16, (size_t)c.nkv * max_seq * hd);
     ALLOC(l.vc, u16, (size_t)c.nkv * max_seq * hd);
     ALLOC(l.vtc, u16, (size_t)c.nkv * hd * max_seq);
+    // zero the caches: positions >= n are masked (weight 0) but 0*NaN from
+    // allocator garbage would still poison the PV MFMA
+    HIP_TRY(hipMemset(l.kc, 0, (size_t)c.nkv * max_seq * hd * 2));
+    HIP_TRY(hipMemset(l.vc, 0, (size_t)c.nkv * max_seq * hd * 2));
+    HIP_TRY(hipMemset(l.vtc, 0, (size_t)c.nkv * hd * max_seq * 2));
   }
   if (e->has_embed()) ALLOC(e->embed, u16, (size_t)V * H);
   if (e->has_head()) {
@@ -948,6 +953,15 @@ extern "C" int cake_hip_reset(cake_engine* e) {
   HIP_TRY(hipMemset(e->dev_pos, 0, 4));
   HIP_TRY(hipMemset(e->dev_step, 0, 4));
   HIP_TRY(hipMemset(e->dev_tok, 0, 4));
+  // clear the KV caches (Goodbye semantics, worker.rs:364-384 /
+  // cache.rs:248-253)
+  const size_t kv_bytes = (size_t)e->c.nkv * e->max_seq * e->c.hd() * 2;
+  for (auto& l : e->L) {
+    HIP_TRY(hipMemsetAsync(l.kc, 0, kv_bytes, e->stream));
+    HIP_TRY(hipMemsetAsync(l.vc, 0, kv_bytes, e->stream));
+    HIP_TRY(hipMemsetAsync(l.vtc, 0, kv_bytes, e->stream));
+  }
+  HIP_TRY(hipStreamSynchronize(e->stream));
   e->host_pos = 0;
   return 0;
 }
