@@ -1,0 +1,201 @@
+"""OpenAI-compatible serving front-end over the MI355X engine.
+
+Mirrors cake's API surface (cake-core/src/cake/sharding/api/mod.rs:66-110):
+  POST /v1/chat/completions   (streaming SSE + non-streaming, api/text.rs)
+  POST /v1/completions
+  GET  /v1/models
+  GET  /api/v1/topology       (api/mod.rs:104-107)
+
+Generation is the same hot loop the bench measures (Master::generate_text,
+master.rs:109-171): greedy ArgMax decode (temperature <= 0 semantics,
+text_model.rs:104); non-greedy sampling is out of round-1 scope and
+`temperature` is accepted but treated as 0.
+
+Tokenization: pass a `tokenizers.Tokenizer` (tokenizer.json) for text
+prompts; without one, requests supply `prompt_token_ids` directly (the
+engine-level contract).  Chat messages are flattened with a minimal
+"role: content" template when no chat template is available.
+
+Run:  python -m cake_amd.serve --model llama3-8b --port 8000
+(random-init weights unless --safetensors is given — there is no network
+for checkpoints in this environment).
+"""
+import argparse
+import json
+import time
+import uuid
+from typing import Optional
+
+import numpy as np
+
+
+def build_prompt(messages):
+    # minimal template; serving real chats needs the model's own chat
+    # template (out of round-1 scope)
+    parts = [f"{m.get('role', 'user')}: {m.get('content', '')}"
+             for m in messages]
+    parts.append("assistant:")
+    return "\n".join(parts)
+
+
+class GenSession:
+    """Thin generation driver over an Engine-compatible object.
+
+    The engine only needs: reset(), prefill(ids)->int, decode(n)->list[int].
+    """
+
+    def __init__(self, engine, eos_ids=(), chunk=8):
+        self.engine = engine
+        self.eos_ids = set(int(e) for e in eos_ids)
+        self.chunk = chunk
+
+    def generate(self, prompt_ids, max_tokens):
+        """Yield token ids, stopping at EOS or max_tokens (the per-token
+        stream Master::generate_text hands its callback, master.rs:125-168)."""
+        self.engine.reset()
+        first = int(self.engine.prefill(np.asarray(prompt_ids,
+                                                   dtype=np.uint32)))
+        produced = 0
+        tok = first
+        while True:
+            yield tok
+            produced += 1
+            if produced >= max_tokens or tok in self.eos_ids:
+                return
+            n = min(self.chunk, max_tokens - produced)
+            toks = self.engine.decode(n)
+            for i, t in enumerate(toks):
+                t = int(t)
+                if i == len(toks) - 1:
+                    tok = t
+                    break
+                yield t
+                produced += 1
+                if produced >= max_tokens or t in self.eos_ids:
+                    return
+
+
+def create_app(engine, model_name="cake-amd", tokenizer=None, eos_ids=(),
+               topology=None):
+    from fastapi import FastAPI, Request
+    from fastapi.responses import JSONResponse, StreamingResponse
+
+    app = FastAPI(title="cake_amd")
+    sess = GenSession(engine, eos_ids=eos_ids)
+
+    def encode(text):
+        if tokenizer is None:
+            raise ValueError(
+                "no tokenizer loaded — pass prompt_token_ids instead")
+        return tokenizer.encode(text).ids
+
+    def decode_tok(tok):
+        if tokenizer is None:
+            return f"<{tok}>"
+        return tokenizer.decode([tok])
+
+    @app.get("/v1/models")
+    def models():  # api/mod.rs route
+        return {"object": "list",
+                "data": [{"id": model_name, "object": "model",
+                          "owned_by": "cake_amd"}]}
+
+    @app.get("/api/v1/topology")
+    def topo():  # api/mod.rs:104-107
+        return topology or {}
+
+    async def run_request(body, kind):
+        max_tokens = int(body.get("max_tokens", 128))
+        stream = bool(body.get("stream", False))
+        if "prompt_token_ids" in body:
+            ids = [int(t) for t in body["prompt_token_ids"]]
+        elif kind == "chat":
+            ids = encode(build_prompt(body.get("messages", [])))
+        else:
+            ids = encode(body.get("prompt", ""))
+        rid = f"chatcmpl-{uuid.uuid4().hex[:12]}"
+        created = int(time.time())
+
+        if stream:
+            def sse():
+                for tok in sess.generate(ids, max_tokens):
+                    delta = ({"content": decode_tok(tok)} if kind == "chat"
+                             else None)
+                    chunk = {
+                        "id": rid, "object": "chat.completion.chunk",
+                        "created": created, "model": model_name,
+                        "choices": [{
+                            "index": 0,
+                            "delta": delta if kind == "chat" else None,
+                            "text": (None if kind == "chat"
+                                     else decode_tok(tok)),
+                            "token_id": int(tok),
+                            "finish_reason": None,
+                        }],
+                    }
+                    yield f"data: {json.dumps(chunk)}\n\n"
+                yield "data: [DONE]\n\n"
+            return StreamingResponse(sse(), media_type="text/event-stream")
+
+        toks = list(sess.generate(ids, max_tokens))
+        text = ("".join(decode_tok(t) for t in toks) if tokenizer
+                else None)
+        finish = "stop" if (toks and toks[-1] in sess.eos_ids) else "length"
+        if kind == "chat":
+            choice = {"index": 0, "finish_reason": finish,
+                      "message": {"role": "assistant", "content": text},
+                      "token_ids": [int(t) for t in toks]}
+        else:
+            choice = {"index": 0, "finish_reason": finish, "text": text,
+                      "token_ids": [int(t) for t in toks]}
+        return JSONResponse({
+            "id": rid, "object": "chat.completion", "created": created,
+            "model": model_name, "choices": [choice],
+            "usage": {"prompt_tokens": len(ids),
+                      "completion_tokens": len(toks),
+                      "total_tokens": len(ids) + len(toks)},
+        })
+
+    @app.post("/v1/chat/completions")
+    async def chat(request: Request):
+        return await run_request(await request.json(), "chat")
+
+    @app.post("/v1/completions")
+    async def completions(request: Request):
+        return await run_request(await request.json(), "completions")
+
+    return app
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--safetensors", default=None)
+    ap.add_argument("--tokenizer", default=None,
+                    help="path to a tokenizer.json")
+    ap.add_argument("--port", type=int, default=8000)
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--max-seq", type=int, default=4096)
+    args = ap.parse_args()
+
+    import cake_amd
+    from cake_amd.configs import MODELS
+    cfg = MODELS[args.model]
+    eng = cake_amd.Engine(json.dumps(cfg), max_seq=args.max_seq,
+                          max_batch_tokens=2048)
+    if args.safetensors:
+        eng.load_safetensors(args.safetensors)
+    else:
+        eng.init_random()
+    tok = None
+    if args.tokenizer:
+        from tokenizers import Tokenizer
+        tok = Tokenizer.from_file(args.tokenizer)
+    app = create_app(eng, model_name=args.model, tokenizer=tok)
+
+    import uvicorn
+    uvicorn.run(app, host=args.host, port=args.port)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,101 @@
+"""API front-end tests with an offline mock engine — the pattern cake's own
+API tests use (api/test_helpers.rs MockTextGenerator + actix test apps,
+SURVEY.md §4).  No GPU needed."""
+import json
+
+import numpy as np
+import pytest
+
+from cake_amd.serve import GenSession, create_app, build_prompt
+
+
+class MockEngine:
+    """Echo-style mock: next token = (last prompt token + step) % 1000."""
+
+    def __init__(self):
+        self.reset_calls = 0
+        self.last = 0
+        self.step = 0
+
+    def reset(self):
+        self.reset_calls += 1
+        self.step = 0
+
+    def prefill(self, ids):
+        self.last = int(ids[-1])
+        self.step = 1
+        return (self.last + 1) % 1000
+
+    def decode(self, n):
+        out = [(self.last + 1 + self.step + i) % 1000 for i in range(n)]
+        self.step += n
+        return np.array(out, dtype=np.uint32)
+
+
+def test_gensession_streams_and_stops_at_max():
+    s = GenSession(MockEngine(), chunk=4)
+    toks = list(s.generate([5, 6, 7], 10))
+    assert len(toks) == 10
+    assert toks[0] == 8
+
+
+def test_gensession_eos_stops():
+    eng = MockEngine()
+    s = GenSession(eng, eos_ids={11}, chunk=4)
+    toks = list(s.generate([5, 6, 7], 50))
+    assert toks[-1] == 11
+    assert len(toks) < 50
+
+
+@pytest.fixture()
+def client():
+    from fastapi.testclient import TestClient
+    app = create_app(MockEngine(), model_name="mock",
+                     topology={"rank0": {"layers": ["model.layers.0-3"]}})
+    return TestClient(app)
+
+
+def test_models_route(client):
+    r = client.get("/v1/models")
+    assert r.status_code == 200
+    assert r.json()["data"][0]["id"] == "mock"
+
+
+def test_topology_route(client):
+    r = client.get("/api/v1/topology")
+    assert r.json()["rank0"]["layers"] == ["model.layers.0-3"]
+
+
+def test_chat_completion_with_token_ids(client):
+    r = client.post("/v1/chat/completions", json={
+        "prompt_token_ids": [1, 2, 3], "max_tokens": 5})
+    assert r.status_code == 200
+    body = r.json()
+    c = body["choices"][0]
+    assert c["token_ids"] == [4, 5, 6, 7, 8]
+    assert c["finish_reason"] == "length"
+    assert body["usage"]["completion_tokens"] == 5
+
+
+def test_chat_completion_streaming(client):
+    with client.stream("POST", "/v1/chat/completions", json={
+            "prompt_token_ids": [9], "max_tokens": 3,
+            "stream": True}) as r:
+        assert r.status_code == 200
+        lines = [ln for ln in r.iter_lines() if ln.startswith("data:")]
+    assert lines[-1] == "data: [DONE]"
+    toks = [json.loads(ln[5:])["choices"][0]["token_id"]
+            for ln in lines[:-1]]
+    assert toks == [10, 11, 12]
+
+
+def test_completions_route(client):
+    r = client.post("/v1/completions", json={
+        "prompt_token_ids": [100], "max_tokens": 2})
+    assert r.json()["choices"][0]["token_ids"] == [101, 102]
+
+
+def test_build_prompt():
+    p = build_prompt([{"role": "system", "content": "a"},
+                      {"role": "user", "content": "b"}])
+    assert p == "system: a\nuser: b\nassistant:"
