@@ -181,6 +181,9 @@ def main():
     def barrier_sync():
         eng.sync()
         if dist:
+            import torch
+            if torch.cuda.is_available():
+                torch.cuda.synchronize(local_rank)
             dist.barrier()
 
     # ---- prefill the prompt + warmup ------------------------------------

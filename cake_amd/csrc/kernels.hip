@@ -994,63 +994,96 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
 #define WS_LOAD(p)                                                         \
   __hip_atomic_load((p), __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)
 
-  float q0 = 0.f, q1 = 0.f;
-  if (act) {
-    q0 = b2f(q[(size_t)h * hd + e0]);
-    q1 = b2f(q[(size_t)h * hd + e0 + 1]);
-  }
-  const float scale = rsqrtf((float)hd);
-  float m = -INFINITY, l = 0.f, o0 = 0.f, o1 = 0.f;
-  const u16* kbase = kc + (size_t)kvh * max_seq * hd;
-  const u16* vbase = vc + (size_t)kvh * max_seq * hd;
-  for (int p = start + wid; p < end; p += 4) {
-    float dot = 0.f;
-    if (act) {
-      const u16* kr = kbase + (size_t)p * hd + e0;
-      dot = q0 * b2f(kr[0]) + q1 * b2f(kr[1]);
-    }
-    dot = wave_sum(dot) * scale;
-    dot = __shfl(dot, 0, WAVE);
-    float mn = fmaxf(m, dot);
-    float alpha = __expf(m - mn);
-    float pw = __expf(dot - mn);
-    float v0 = 0.f, v1 = 0.f;
-    if (act) {
-      const u16* vr = vbase + (size_t)p * hd + e0;
-      v0 = b2f(vr[0]);
-      v1 = b2f(vr[1]);
-    }
-    o0 = o0 * alpha + pw * v0;
-    o1 = o1 * alpha + pw * v1;
-    l = l * alpha + pw;
-    m = mn;
-  }
-  // combine the 4 waves' partials
+  // Tiled two-phase structure: the serial online-softmax chain runs once
+  // per TILE (128 positions), not per position — phase A computes a tile's
+  // scores with 16-B coalesced K loads (16 lanes per position), phase B is
+  // a block-wide softmax step, phase C accumulates PV with coalesced,
+  // mutually independent V loads.  The per-position serial chain of the
+  // naive version measured 128 GB/s at n=2048.
   __shared__ float sm[4], sl[4];
   __shared__ float so[4][128 + 8];
-  if (lane == 0) { sm[wid] = m; sl[wid] = l; }
+  __shared__ float stile[128];
+  const float scale = rsqrtf((float)hd);
+  const u16* kbase = kc + (size_t)kvh * max_seq * hd;
+  const u16* vbase = vc + (size_t)kvh * max_seq * hd;
+  // phase-A per-thread q slice: dims (t&15)*8 .. +8
+  const int dgrp = t & 15;
+  float qa[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int d = dgrp * 8 + j;
+    qa[j] = d < hd ? b2f(q[(size_t)h * hd + d]) : 0.f;
+  }
+  float m = -INFINITY, l = 0.f, o0 = 0.f, o1 = 0.f;
+  const int TILE = 128;
+  for (int sub0 = start; sub0 < end; sub0 += TILE) {
+    // --- phase A: scores for [sub0, sub0+TILE) --------------------------
+    // 256 threads = 16 positions per pass (16 lanes per position, each
+    // loading 16 B of the K row)
+#pragma unroll 2
+    for (int pass = 0; pass < TILE / 16; ++pass) {
+      const int p = sub0 + pass * 16 + (t >> 4);
+      float d = 0.f;
+      if (p < end && dgrp * 8 < hd) {
+        const u16* kr = kbase + (size_t)p * hd + dgrp * 8;
+        short8 kv8 = *reinterpret_cast<const short8*>(kr);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) d = fmaf(b2f((u16)kv8[j]), qa[j], d);
+      }
+      // reduce across the 16 lanes of this position
+#pragma unroll
+      for (int off = 8; off >= 1; off >>= 1) d += __shfl_xor(d, off, 16);
+      if (dgrp == 0)
+        stile[pass * 16 + (t >> 4)] = (p < end) ? d * scale : -INFINITY;
+    }
+    __syncthreads();
+    // --- phase B: block softmax step over the tile ----------------------
+    float lm = -INFINITY;
+    for (int i = t; i < TILE; i += blockDim.x) lm = fmaxf(lm, stile[i]);
+    lm = wave_max(lm);
+    if (lane == 0) sm[wid] = lm;
+    __syncthreads();
+    const float tmax = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
+    const float mnew = fmaxf(m, tmax);
+    const float alpha = (mnew == -INFINITY) ? 0.f : __expf(m - mnew);
+    float psum = 0.f;
+    for (int i = t; i < TILE; i += blockDim.x) {
+      const float sv = stile[i];
+      const float e = (sv == -INFINITY) ? 0.f : __expf(sv - mnew);
+      stile[i] = e;
+      psum += e;
+    }
+    psum = wave_sum(psum);
+    if (lane == 0) sl[wid] = psum;
+    __syncthreads();
+    l = l * alpha + sl[0] + sl[1] + sl[2] + sl[3];
+    o0 *= alpha;
+    o1 *= alpha;
+    m = mnew;
+    // --- phase C: PV accumulate (thread t: dims 2*lane, position residue
+    // wid mod 4; loads independent across iterations) --------------------
+#pragma unroll 4
+    for (int k = 0; k < TILE / 4; ++k) {
+      const int po = k * 4 + wid;
+      const int p = sub0 + po;
+      if (act && p < end) {
+        const float w = stile[po];
+        const u16* vr = vbase + (size_t)p * hd + e0;
+        o0 = fmaf(w, b2f(vr[0]), o0);
+        o1 = fmaf(w, b2f(vr[1]), o1);
+      }
+    }
+    __syncthreads();  // stile reused next sub-tile
+  }
+  // combine the 4 position-residue partials per dim
   if (act) { so[wid][e0] = o0; so[wid][e0 + 1] = o1; }
   __syncthreads();
   if (t == 0) {
-    float M = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
-    float L = 0.f;
-#pragma unroll
-    for (int w = 0; w < 4; ++w)
-      L += (sm[w] == -INFINITY) ? 0.f : sl[w] * __expf(sm[w] - M);
-    WS_STORE(&wsrow[hd], M);
-    WS_STORE(&wsrow[hd + 1], L);
+    WS_STORE(&wsrow[hd], m);
+    WS_STORE(&wsrow[hd + 1], l);
   }
-  __syncthreads();
-  {
-    float M = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
-    for (int d = t; d < hd; d += blockDim.x) {
-      float o = 0.f;
-#pragma unroll
-      for (int w = 0; w < 4; ++w)
-        o += (sm[w] == -INFINITY) ? 0.f : so[w][d] * __expf(sm[w] - M);
-      WS_STORE(&wsrow[d], o);
-    }
-  }
+  for (int d = t; d < hd; d += blockDim.x)
+    WS_STORE(&wsrow[d], so[0][d] + so[1][d] + so[2][d] + so[3][d]);
 
   // ---- publish partial + elect the combining block ----------------------
   // sc1 write-through publish (Guideline 16 R1 variant): the ws stores above
