@@ -1020,7 +1020,7 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
     // --- phase A: scores for [sub0, sub0+TILE) --------------------------
     // 256 threads = 16 positions per pass (16 lanes per position, each
     // loading 16 B of the K row)
-#pragma unroll 2
+#pragma unroll 4
     for (int pass = 0; pass < TILE / 16; ++pass) {
       const int p = sub0 + pass * 16 + (t >> 4);
       float d = 0.f;
@@ -1062,7 +1062,7 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
     m = mnew;
     // --- phase C: PV accumulate (thread t: dims 2*lane, position residue
     // wid mod 4; loads independent across iterations) --------------------
-#pragma unroll 4
+#pragma unroll 8
     for (int k = 0; k < TILE / 4; ++k) {
       const int po = k * 4 + wid;
       const int p = sub0 + po;
