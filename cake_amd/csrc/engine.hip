@@ -628,7 +628,7 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
     int v = atoi(gr);
     if (v == 4 || v == 8) e->gu_rows = v;
   }
-  ALLOC(e->attn_ws, float, (size_t)c.nh * e->nchunk * (hd + 2));
+  ALLOC(e->attn_ws, float, (size_t)c.nh * 64 * (hd + 2));  // nchunk <= 64
   ALLOC(e->attn_cnt, u32, c.nh);
   HIP_TRY(hipMemset(e->attn_cnt, 0, sizeof(u32) * c.nh));
   ALLOC(e->dev_pos, int, 1);
@@ -1035,6 +1035,10 @@ extern "C" int cake_hip_decode(cake_engine* e, int steps,
   if (e->host_pos + steps > e->max_seq)
     return set_err(5, "decode exceeds max_seq");
   HIP_TRY(hipMemsetAsync(e->dev_step, 0, 4, e->stream));
+  // pick the split-KV chunk count for the expected context length (fixed
+  // once a graph is captured; CAKE_NCHUNK overrides)
+  if (!getenv("CAKE_NCHUNK") && !e->graph)
+    e->nchunk = e->host_pos >= 1024 ? 16 : 8;
   bool graph_ok = e->use_graph() && e->world == 1 && !e->st.on;
   if (graph_ok && !e->graph) {
     // capture one decode step (device-side token/pos/ring make it replayable)
