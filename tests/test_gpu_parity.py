@@ -101,6 +101,8 @@ def test_op_gemv_small_k():
     (200, 260, 128),       # M and N tails
     (2048, 1536, 4096),    # prefill-like shape
     (33, 128256 // 16, 64),  # wide-N-ish
+    (2048, 28672, 4096),   # gate_up prefill shape (256^2 counted-vmcnt path)
+    (300, 51300, 256),     # 256^2 path with M and N tails
 ])
 def test_op_gemm(M, N, K):
     # asymmetric random operands (transpose-detecting — guide §5.4 rule 16)
