@@ -338,6 +338,43 @@ def eng2_greedy(eng, prompt, n):
     return [first] + list(eng.decode(n - 1))
 
 
+def test_chunked_prefill_parity():
+    """Prompts longer than max_batch_tokens prefill in chunks (the engine's
+    pos0-offset path, mirroring cake's whole-prompt forward in pieces):
+    results must equal a single-chunk prefill and the oracle."""
+    import tempfile, os
+    cfg_json = dict(
+        model_type="llama", hidden_size=256, intermediate_size=512,
+        vocab_size=512, num_hidden_layers=2, num_attention_heads=2,
+        num_key_value_heads=1, head_dim=128, rms_norm_eps=1e-5,
+        rope_theta=500000.0, max_position_embeddings=1024,
+        tie_word_embeddings=False)
+    cfg = Config.from_json(cfg_json)
+    w = random_weights(cfg, seed=41)
+    oracle = quantized_oracle(cfg, w)
+    with tempfile.TemporaryDirectory() as td:
+        st = os.path.join(td, "m.safetensors")
+        weights_to_safetensors(w, cfg, st)
+        big = cake_amd.Engine(json.dumps(cfg_json), max_seq=512,
+                              max_batch_tokens=512)
+        big.load_safetensors(st)
+        small = cake_amd.Engine(json.dumps(cfg_json), max_seq=512,
+                                max_batch_tokens=48)  # forces 48+48+4 chunks
+        small.load_safetensors(st)
+        try:
+            rng = np.random.default_rng(17)
+            prompt = rng.integers(0, cfg.vocab_size, size=100).astype(
+                np.uint32)
+            _, l_big = big.prefill(prompt, want_logits=True)
+            _, l_small = small.prefill(prompt, want_logits=True)
+            assert rel_err(l_small, l_big) < 1e-3, "chunked != single-chunk"
+            ref = oracle.forward(prompt[None, :].astype(np.int64), 0)[0]
+            assert rel_err(l_small, ref) < 2e-2
+        finally:
+            big.close()
+            small.close()
+
+
 def test_long_context_decode_parity():
     """Multi-sub-tile decode attention (the tiled split-KV kernel's rescale
     across 128-position sub-tiles): with CAKE_NCHUNK=2 and a 300-token

@@ -93,3 +93,19 @@ def test_engine_requires_gpu_no_silent_fallback():
             model_type="llama", hidden_size=64, intermediate_size=128,
             vocab_size=128, num_hidden_layers=1, num_attention_heads=4,
             num_key_value_heads=2, head_dim=16), max_seq=32)
+
+
+def test_engine_rejects_bad_config():
+    # config parsing precedes any GPU call, so these are CPU-testable
+    with pytest.raises(CakeHipError, match="config.json"):
+        cake_amd.Engine("not json at all {", max_seq=32)
+    with pytest.raises(CakeHipError, match="missing required"):
+        cake_amd.Engine(dict(model_type="llama", hidden_size=64), max_seq=32)
+
+
+def test_engine_rejects_bad_layer_range():
+    with pytest.raises(CakeHipError, match="layer range"):
+        cake_amd.Engine(dict(
+            model_type="llama", hidden_size=64, intermediate_size=128,
+            vocab_size=128, num_hidden_layers=2, num_attention_heads=4,
+            num_key_value_heads=2, head_dim=16), layer_lo=1, layer_hi=1)
