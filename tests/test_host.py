@@ -97,10 +97,12 @@ def test_engine_requires_gpu_no_silent_fallback():
 
 def test_engine_rejects_bad_config():
     # config parsing precedes any GPU call, so these are CPU-testable
-    with pytest.raises(CakeHipError, match="config.json"):
-        cake_amd.Engine("not json at all {", max_seq=32)
+    import json as _json
+    with pytest.raises(_json.JSONDecodeError):
+        cake_amd.Engine("not json at all {", max_seq=32)  # wrapper-side
     with pytest.raises(CakeHipError, match="missing required"):
-        cake_amd.Engine(dict(model_type="llama", hidden_size=64), max_seq=32)
+        cake_amd.Engine(dict(model_type="llama", hidden_size=64,
+                             num_hidden_layers=1), max_seq=32)
 
 
 def test_engine_rejects_bad_layer_range():
