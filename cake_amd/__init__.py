@@ -171,8 +171,8 @@ class Engine:
             flags = HAS_EMBED | HAS_HEAD | USE_GRAPH
         self.cfg = cfg
         self.flags = flags
-        self.vocab = cfg["vocab_size"]
-        self.hidden = cfg["hidden_size"]
+        self.vocab = cfg.get("vocab_size", 0)
+        self.hidden = cfg.get("hidden_size", 0)
         h = ctypes.c_void_p()
         _check(_lib.cake_hip_engine_create(
             config.encode(), layer_lo, layer_hi, flags, max_seq,
