@@ -1844,7 +1844,11 @@ void launch_gemv_fp8(const unsigned char* W, const float* sc, const u16* x,
                      void* out, const u16* res, const u16* nw, float eps,
                      int N, int K, int epi, hipStream_t s) {
   const int nkb = (K + 127) / 128;
-  const int rows = N >= 16384 ? 8 : 4;
+  static const int env_rows = [] {
+    const char* v = getenv("CAKE_FP8_ROWS");
+    return v ? atoi(v) : 0;
+  }();
+  const int rows = N >= 16384 ? 8 : (env_rows ? env_rows : 4);
 #define F8_KB(R, EPI, KB)                                                   \
   do {                                                                      \
     dim3 grid((N + R - 1) / R);                                             \
