@@ -1,0 +1,425 @@
+"""cake wire-protocol interop: a worker that an UNMODIFIED cake master can
+drive over TCP (SURVEY.md §8f item 4 — the literal "drops in as a backend"
+story).
+
+Framing (proto/mod.rs:4-10 + proto/message.rs:330-394): 8-byte header =
+u64 BE of (PROTO_MAGIC << 32 | payload_len), payload <= 512 MiB, body =
+speedy 0.8.7 BigEndian encoding of the `Message` enum
+(proto/message.rs:191-247).
+
+speedy 0.8.7 BigEndian encoding, restated for the types on this wire:
+  - enum: u32 BE variant tag, declaration order from 0
+  - struct / tuple: fields in declaration order
+  - Vec<T> / String: u32 BE length, then elements / UTF-8 bytes
+  - u8: 1 byte; u32: 4 BE; u64: 8 BE; u128: 16 BE; bool: 1 byte
+  - usize: serialized as u64 BE
+PARITY NOTE: cake itself cannot be compiled in this container (no Rust
+toolchain — SURVEY.md §8c), so this byte layout is pinned by restatement of
+speedy's published format plus self-roundtrip tests (tests/test_wire.py),
+not by captured reference bytes; byte-level parity vs a live cake master is
+"partial" until verified against one.
+
+RawTensor (proto/message.rs:39-47): { data: Vec<u8>, dtype: u8,
+shape: Vec<usize> }, dtype tags per dtype_to_u8 (message.rs:8-21).
+
+The worker loop mirrors `handle_master_client` (sharding/worker.rs:299-578):
+Hello -> WorkerInfo; then SingleOp/Batch -> forward this shard's layers ->
+Tensor reply; Goodbye -> clear KV; errors -> WorkerError replies
+(worker.rs:490-503).  PSK auth (auth.rs) is not implemented (out of round-1
+scope; run cake without --cluster-key).
+
+Run:  python -m cake_amd.wire --model llama3-8b --layers 0-15 --port 10128
+"""
+import asyncio
+import struct
+
+import numpy as np
+
+PROTO_MAGIC = 0x104F4C7
+MESSAGE_MAX_SIZE = 512 * 1024 * 1024
+
+# dtype tags (proto/message.rs:8-21)
+DT_U8, DT_U32, DT_I64, DT_BF16, DT_F16, DT_F32, DT_F64, DT_F8E4M3 = range(8)
+
+# Message variant tags (declaration order, proto/message.rs:191-247)
+MSG_HELLO = 0
+MSG_WORKER_INFO = 1
+MSG_SINGLE_OP = 2
+MSG_BATCH = 3
+MSG_TENSOR = 4
+MSG_GOODBYE = 5
+MSG_LAYER_ASSIGNMENT = 6
+MSG_LAYER_ASSIGNMENT_ACK = 7
+MSG_MODEL_DATA_CHUNK = 8
+MSG_MODEL_DATA_DONE = 9
+MSG_MODEL_DATA_RESUME = 10
+MSG_WORKER_READY = 11
+MSG_WORKER_ERROR = 12
+
+
+class Writer:
+    def __init__(self):
+        self.b = bytearray()
+
+    def u8(self, v):
+        self.b.append(v & 0xFF)
+
+    def u32(self, v):
+        self.b += struct.pack(">I", v)
+
+    def u64(self, v):
+        self.b += struct.pack(">Q", v)
+
+    def u128(self, v):
+        self.b += v.to_bytes(16, "big")
+
+    def usize(self, v):
+        self.u64(v)
+
+    def string(self, s):
+        raw = s.encode()
+        self.u32(len(raw))
+        self.b += raw
+
+    def vec_u8(self, data):
+        self.u32(len(data))
+        self.b += data
+
+    def vec_usize(self, xs):
+        self.u32(len(xs))
+        for x in xs:
+            self.usize(x)
+
+
+class Reader:
+    def __init__(self, b):
+        self.b = b
+        self.o = 0
+
+    def take(self, n):
+        if self.o + n > len(self.b):
+            raise ValueError("wire: truncated message")
+        v = self.b[self.o:self.o + n]
+        self.o += n
+        return v
+
+    def u8(self):
+        return self.take(1)[0]
+
+    def u32(self):
+        return struct.unpack(">I", self.take(4))[0]
+
+    def u64(self):
+        return struct.unpack(">Q", self.take(8))[0]
+
+    def u128(self):
+        return int.from_bytes(self.take(16), "big")
+
+    def usize(self):
+        return self.u64()
+
+    def string(self):
+        return self.take(self.u32()).decode()
+
+    def vec_u8(self):
+        return bytes(self.take(self.u32()))
+
+    def vec_usize(self):
+        return [self.usize() for _ in range(self.u32())]
+
+
+# ---------------------------------------------------------------------------
+# RawTensor <-> numpy
+# ---------------------------------------------------------------------------
+def raw_tensor_to_numpy(dtype, shape, data):
+    """Decode wire bytes; bf16 is widened to f32 (numpy has no bf16).
+    Byte order on the wire is the tensor's native (little-endian) layout —
+    speedy serializes Vec<u8> verbatim (RawTensor::from_tensor memcpy)."""
+    if dtype == DT_F32:
+        a = np.frombuffer(data, dtype="<f4")
+    elif dtype == DT_BF16:
+        u = np.frombuffer(data, dtype="<u2").astype(np.uint32) << 16
+        a = u.view(np.float32)
+    elif dtype == DT_F16:
+        a = np.frombuffer(data, dtype="<f2").astype(np.float32)
+    elif dtype == DT_U32:
+        a = np.frombuffer(data, dtype="<u4")
+    elif dtype == DT_I64:
+        a = np.frombuffer(data, dtype="<i8")
+    elif dtype == DT_F64:
+        a = np.frombuffer(data, dtype="<f8")
+    elif dtype == DT_U8:
+        a = np.frombuffer(data, dtype=np.uint8)
+    else:
+        raise ValueError(f"wire: unsupported dtype tag {dtype}")
+    return a.reshape(shape)
+
+
+def numpy_to_raw(a, dtype):
+    """Encode f32 numpy as the requested wire dtype."""
+    if dtype == DT_F32:
+        data = np.ascontiguousarray(a, dtype="<f4").tobytes()
+    elif dtype == DT_BF16:
+        # RNE f32 -> bf16 (matches the device conversion)
+        u = np.ascontiguousarray(a, dtype=np.float32).view(np.uint32)
+        r = u + 0x7FFF + ((u >> 16) & 1)
+        nan = (u & 0x7FFFFFFF) > 0x7F800000
+        r = np.where(nan, np.uint32(0x7FC00000), r)
+        data = (r >> 16).astype("<u2").tobytes()
+    elif dtype == DT_F16:
+        data = np.ascontiguousarray(a, dtype="<f2").tobytes()
+    else:
+        raise ValueError(f"wire: unsupported reply dtype {dtype}")
+    return dtype, list(a.shape), data
+
+
+def write_raw_tensor(w, dtype, shape, data):
+    w.vec_u8(data)
+    w.u8(dtype)
+    w.vec_usize(shape)
+
+
+def read_raw_tensor(r):
+    data = r.vec_u8()
+    dtype = r.u8()
+    shape = r.vec_usize()
+    return dtype, shape, data
+
+
+# ---------------------------------------------------------------------------
+# Message encode/decode (the subset the worker loop speaks)
+# ---------------------------------------------------------------------------
+def encode_message(msg):
+    """msg: dict with 'type' plus fields."""
+    w = Writer()
+    t = msg["type"]
+    w.u32(t)
+    if t in (MSG_HELLO, MSG_GOODBYE, MSG_MODEL_DATA_DONE, MSG_WORKER_READY):
+        pass
+    elif t == MSG_WORKER_INFO:
+        i = msg["info"]
+        w.string(i["version"])
+        w.string(i["dtype"])
+        w.string(i["os"])
+        w.string(i["arch"])
+        w.string(i["device"])
+        w.usize(i["device_idx"])
+        w.u128(i["latency"])
+    elif t == MSG_SINGLE_OP:
+        w.string(msg["layer_name"])
+        write_raw_tensor(w, *msg["x"])
+        w.usize(msg["index_pos"])
+        w.usize(msg["block_idx"])
+    elif t == MSG_BATCH:
+        write_raw_tensor(w, *msg["x"])
+        w.u32(len(msg["batch"]))
+        for name, ip, bi in msg["batch"]:
+            w.string(name)
+            w.usize(ip)
+            w.usize(bi)
+    elif t == MSG_TENSOR:
+        write_raw_tensor(w, *msg["x"])
+    elif t == MSG_WORKER_ERROR:
+        w.string(msg["message"])
+    else:
+        raise ValueError(f"wire: cannot encode message type {t}")
+    return bytes(w.b)
+
+
+def decode_message(buf):
+    r = Reader(buf)
+    t = r.u32()
+    out = {"type": t}
+    if t in (MSG_HELLO, MSG_GOODBYE, MSG_MODEL_DATA_DONE, MSG_WORKER_READY):
+        pass
+    elif t == MSG_WORKER_INFO:
+        out["info"] = dict(
+            version=r.string(), dtype=r.string(), os=r.string(),
+            arch=r.string(), device=r.string(), device_idx=r.usize(),
+            latency=r.u128())
+    elif t == MSG_SINGLE_OP:
+        out["layer_name"] = r.string()
+        out["x"] = read_raw_tensor(r)
+        out["index_pos"] = r.usize()
+        out["block_idx"] = r.usize()
+    elif t == MSG_BATCH:
+        out["x"] = read_raw_tensor(r)
+        n = r.u32()
+        out["batch"] = [(r.string(), r.usize(), r.usize()) for _ in range(n)]
+    elif t == MSG_TENSOR:
+        out["x"] = read_raw_tensor(r)
+    elif t == MSG_WORKER_ERROR:
+        out["message"] = r.string()
+    else:
+        raise ValueError(f"wire: cannot decode message type {t}")
+    return out
+
+
+def frame(payload: bytes) -> bytes:
+    if len(payload) > MESSAGE_MAX_SIZE:
+        raise ValueError("wire: message too large")
+    return struct.pack(">Q", (PROTO_MAGIC << 32) | len(payload)) + payload
+
+
+async def read_framed(reader) -> bytes:
+    header = await reader.readexactly(8)
+    h = struct.unpack(">Q", header)[0]
+    magic, size = h >> 32, h & 0xFFFFFFFF
+    if magic != PROTO_MAGIC:
+        raise ValueError(f"invalid magic value: {magic}")
+    if size > MESSAGE_MAX_SIZE:
+        raise ValueError(f"request size {size} > MESSAGE_MAX_SIZE")
+    return await reader.readexactly(size)
+
+
+# ---------------------------------------------------------------------------
+# The worker (handle_master_client mirror, worker.rs:299-578)
+# ---------------------------------------------------------------------------
+class WireWorker:
+    """Serves this shard's layers over cake's wire protocol.
+
+    `engine` needs: forward_hidden(x_f32 (S,H), index_pos) -> (S,H) f32,
+    reset().  `layer_names` is the ordered list of layer names this worker
+    owns (e.g. model.layers.4 .. model.layers.7)."""
+
+    def __init__(self, engine, layer_names, dtype="BF16", device="gfx950"):
+        self.engine = engine
+        self.layer_names = list(layer_names)
+        self.dtype_name = dtype
+        self.device = device
+
+    def worker_info(self):
+        import platform
+        return dict(version="0.1.0-mi355x", dtype=self.dtype_name,
+                    os=platform.system().lower(), arch=platform.machine(),
+                    device=self.device, device_idx=0, latency=0)
+
+    def _forward(self, msg):
+        dtype, shape, data = msg["x"]
+        x = raw_tensor_to_numpy(dtype, shape, data).astype(np.float32)
+        if x.ndim == 3 and x.shape[0] == 1:
+            x2 = x[0]
+        elif x.ndim == 2:
+            x2 = x
+        else:
+            raise ValueError(f"unexpected activation shape {shape}")
+        if msg["type"] == MSG_BATCH:
+            names = [b[0] for b in msg["batch"]]
+            index_pos = msg["batch"][0][1]
+        else:
+            names = [msg["layer_name"]]
+            index_pos = msg["index_pos"]
+        # the engine runs its whole contiguous shard; the batch must be it
+        if names != self.layer_names[:len(names)] or \
+                len(names) != len(self.layer_names):
+            raise ValueError(
+                f"batch {names} != this worker's layers {self.layer_names}")
+        y = self.engine.forward_hidden(x2, index_pos)
+        out = y.reshape(x.shape)
+        return numpy_to_raw(out, dtype)
+
+    async def handle(self, reader, writer):
+        try:
+            while True:
+                try:
+                    payload = await read_framed(reader)
+                except (asyncio.IncompleteReadError, ConnectionError):
+                    break
+                msg = decode_message(payload)
+                t = msg["type"]
+                if t == MSG_HELLO:
+                    reply = {"type": MSG_WORKER_INFO,
+                             "info": self.worker_info()}
+                elif t in (MSG_BATCH, MSG_SINGLE_OP):
+                    try:
+                        reply = {"type": MSG_TENSOR, "x": self._forward(msg)}
+                    except Exception as e:  # worker.rs:490-503
+                        reply = {"type": MSG_WORKER_ERROR, "message": str(e)}
+                elif t == MSG_GOODBYE:
+                    self.engine.reset()  # worker.rs:364-384
+                    reply = {"type": MSG_GOODBYE}
+                else:
+                    reply = {"type": MSG_WORKER_ERROR,
+                             "message": f"unsupported message type {t}"}
+                writer.write(frame(encode_message(reply)))
+                await writer.drain()
+                if t == MSG_GOODBYE:
+                    break
+        finally:
+            writer.close()
+
+    async def serve(self, host="0.0.0.0", port=10128):
+        server = await asyncio.start_server(self.handle, host, port)
+        async with server:
+            await server.serve_forever()
+
+
+class WireClient:
+    """Master-side hop — mirrors `Client` (sharding/client.rs:79-174):
+    Hello/WorkerInfo handshake, then blocking Batch -> Tensor per forward."""
+
+    def __init__(self, host, port, dtype=DT_BF16):
+        self.host, self.port, self.dtype = host, port, dtype
+        self.reader = self.writer = None
+        self.info = None
+
+    async def connect(self):
+        self.reader, self.writer = await asyncio.open_connection(
+            self.host, self.port)
+        self.info = (await self._call({"type": MSG_HELLO}))["info"]
+
+    async def _call(self, msg):
+        self.writer.write(frame(encode_message(msg)))
+        await self.writer.drain()
+        return decode_message(await read_framed(self.reader))
+
+    async def forward_batch(self, x, batch):
+        """x: (1, S, H) f32 numpy; batch: [(layer_name, index_pos,
+        block_idx)] — returns (1, S, H) f32 (client.rs:165-174)."""
+        r = await self._call({"type": MSG_BATCH,
+                              "x": numpy_to_raw(x, self.dtype),
+                              "batch": batch})
+        if r["type"] == MSG_WORKER_ERROR:
+            raise RuntimeError(f"worker error: {r['message']}")
+        return raw_tensor_to_numpy(*r["x"]).astype(np.float32)
+
+    async def goodbye(self):
+        await self._call({"type": MSG_GOODBYE})
+        self.writer.close()
+
+
+def main():
+    import argparse
+    import json
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--layers", default=None,
+                    help="A-B inclusive range, e.g. 0-15 (default: all)")
+    ap.add_argument("--safetensors", default=None)
+    ap.add_argument("--port", type=int, default=10128)
+    ap.add_argument("--host", default="0.0.0.0")
+    ap.add_argument("--max-seq", type=int, default=4096)
+    args = ap.parse_args()
+
+    import cake_amd
+    from cake_amd.configs import MODELS
+    cfg = MODELS[args.model]
+    if args.layers:
+        lo, hi = args.layers.split("-")
+        lo, hi = int(lo), int(hi) + 1
+    else:
+        lo, hi = 0, cfg["num_hidden_layers"]
+    eng = cake_amd.Engine(json.dumps(cfg), layer_lo=lo, layer_hi=hi, flags=0,
+                          max_seq=args.max_seq, max_batch_tokens=2048)
+    if args.safetensors:
+        eng.load_safetensors(args.safetensors)
+    else:
+        eng.init_random()
+    names = [f"model.layers.{i}" for i in range(lo, hi)]
+    print(f"[wire] serving {names[0]}..{names[-1]} on "
+          f"{args.host}:{args.port}")
+    asyncio.run(WireWorker(eng, names).serve(args.host, args.port))
+
+
+if __name__ == "__main__":
+    main()

@@ -506,3 +506,63 @@ def test_full_size_8b_determinism_and_graph_parity():
     finally:
         a.close()
         b.close()
+
+
+def test_wire_worker_over_tcp_matches_local():
+    """The literal drop-in story (§8f item 4): the model's second half runs
+    in a wire worker behind cake's TCP protocol; the master-side chain
+    (local shard -> WireClient Batch hop) must match the monolithic engine."""
+    import asyncio
+    import tempfile, os, threading
+    from cake_amd import wire
+
+    cfg_json = dict(
+        model_type="llama", hidden_size=128, intermediate_size=256,
+        vocab_size=256, num_hidden_layers=4, num_attention_heads=4,
+        num_key_value_heads=2, head_dim=32, rms_norm_eps=1e-5,
+        rope_theta=10000.0, max_position_embeddings=128,
+        tie_word_embeddings=False)
+    cfg = Config.from_json(cfg_json)
+    w = random_weights(cfg, seed=55)
+    with tempfile.TemporaryDirectory() as td:
+        st = os.path.join(td, "m.safetensors")
+        weights_to_safetensors(w, cfg, st)
+        full = cake_amd.Engine(json.dumps(cfg_json), max_seq=64,
+                               max_batch_tokens=32,
+                               flags=cake_amd.HAS_EMBED | cake_amd.HAS_HEAD)
+        full.load_safetensors(st)
+        lo_eng = cake_amd.Engine(json.dumps(cfg_json), 0, 2, flags=0,
+                                 max_seq=64, max_batch_tokens=32)
+        lo_eng.load_safetensors(st)
+        hi_eng = cake_amd.Engine(json.dumps(cfg_json), 2, 4, flags=0,
+                                 max_seq=64, max_batch_tokens=32)
+        hi_eng.load_safetensors(st)
+        try:
+            names = ["model.layers.2", "model.layers.3"]
+            worker = wire.WireWorker(hi_eng, names)
+
+            async def run():
+                server = await asyncio.start_server(worker.handle,
+                                                    "127.0.0.1", 0)
+                port = server.sockets[0].getsockname()[1]
+                cli = wire.WireClient("127.0.0.1", port)
+                await cli.connect()
+                assert cli.info["device"] == "gfx950"
+                rng = np.random.default_rng(7)
+                S = 9
+                x = (rng.standard_normal((S, cfg.hidden_size)) * 0.05
+                     ).astype(np.float32)
+                ref = full.forward_hidden(x, 0)
+                a = lo_eng.forward_hidden(x, 0)
+                b = await cli.forward_batch(
+                    a[None], [(n, 0, 2 + i) for i, n in enumerate(names)])
+                assert rel_err(b[0], ref) < 2e-2
+                await cli.goodbye()
+                server.close()
+                await server.wait_closed()
+
+            asyncio.run(asyncio.wait_for(run(), timeout=120))
+        finally:
+            full.close()
+            lo_eng.close()
+            hi_eng.close()

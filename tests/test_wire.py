@@ -1,0 +1,139 @@
+"""cake wire-protocol tests — mirrors the reference's own protocol tests
+(tests/protocol.rs: mock echo worker on 127.0.0.1, byte-exact roundtrips;
+SURVEY.md §4).  CPU-only."""
+import asyncio
+import struct
+
+import numpy as np
+import pytest
+
+from cake_amd import wire
+
+
+def rt(msg):
+    return wire.decode_message(wire.encode_message(msg))
+
+
+def test_simple_messages_roundtrip():
+    for t in (wire.MSG_HELLO, wire.MSG_GOODBYE, wire.MSG_WORKER_READY):
+        assert rt({"type": t})["type"] == t
+
+
+def test_worker_info_roundtrip():
+    info = dict(version="0.1.0", dtype="BF16", os="linux", arch="x86_64",
+                device="gfx950", device_idx=3, latency=12345678901234567890)
+    m = rt({"type": wire.MSG_WORKER_INFO, "info": info})
+    assert m["info"] == info
+
+
+def test_tensor_roundtrip_f32_and_bf16():
+    rng = np.random.default_rng(0)
+    a = rng.standard_normal((2, 5, 64)).astype(np.float32)
+    raw = wire.numpy_to_raw(a, wire.DT_F32)
+    m = rt({"type": wire.MSG_TENSOR, "x": raw})
+    b = wire.raw_tensor_to_numpy(*m["x"])
+    assert np.array_equal(a, b)
+    raw16 = wire.numpy_to_raw(a, wire.DT_BF16)
+    m16 = rt({"type": wire.MSG_TENSOR, "x": raw16})
+    b16 = wire.raw_tensor_to_numpy(*m16["x"])
+    from tests.helpers import quantize_bf16
+    assert np.array_equal(quantize_bf16(a), b16)
+
+
+def test_batch_roundtrip_layout():
+    x = wire.numpy_to_raw(np.ones((1, 1, 8), np.float32), wire.DT_BF16)
+    batch = [("model.layers.4", 17, 4), ("model.layers.5", 17, 5)]
+    m = rt({"type": wire.MSG_BATCH, "x": x, "batch": batch})
+    assert m["batch"] == batch
+    # spot-check the byte layout of the header fields (speedy BE restatement)
+    raw = wire.encode_message({"type": wire.MSG_BATCH, "x": x,
+                               "batch": batch})
+    assert raw[:4] == struct.pack(">I", wire.MSG_BATCH)   # u32 BE tag
+    assert raw[4:8] == struct.pack(">I", 16)              # Vec<u8> len (bf16)
+
+
+def test_framing():
+    payload = wire.encode_message({"type": wire.MSG_HELLO})
+    f = wire.frame(payload)
+    h = struct.unpack(">Q", f[:8])[0]
+    assert h >> 32 == wire.PROTO_MAGIC
+    assert (h & 0xFFFFFFFF) == len(payload)
+
+
+class EchoEngine:
+    """Mock: adds 1.0 to every activation (distinguishes from pure echo)."""
+
+    def __init__(self, n_layers=2):
+        self.resets = 0
+
+    def forward_hidden(self, x, index_pos):
+        return x + 1.0
+
+    def reset(self):
+        self.resets += 1
+
+
+@pytest.mark.parametrize("dtype", [wire.DT_F32, wire.DT_BF16])
+def test_loopback_worker_batch(dtype):
+    async def run():
+        eng = EchoEngine()
+        names = ["model.layers.0", "model.layers.1"]
+        worker = wire.WireWorker(eng, names)
+        server = await asyncio.start_server(worker.handle, "127.0.0.1", 0)
+        port = server.sockets[0].getsockname()[1]
+
+        reader, writer = await asyncio.open_connection("127.0.0.1", port)
+
+        async def call(msg):
+            writer.write(wire.frame(wire.encode_message(msg)))
+            await writer.drain()
+            return wire.decode_message(await wire.read_framed(reader))
+
+        # handshake (worker.rs:313-348)
+        r = await call({"type": wire.MSG_HELLO})
+        assert r["type"] == wire.MSG_WORKER_INFO
+        assert r["info"]["device"] == "gfx950"
+
+        # batch over both layers
+        x = np.arange(16, dtype=np.float32).reshape(1, 2, 8) * 0.25
+        r = await call({"type": wire.MSG_BATCH,
+                        "x": wire.numpy_to_raw(x, dtype),
+                        "batch": [(n, 7, i) for i, n in enumerate(names)]})
+        assert r["type"] == wire.MSG_TENSOR
+        y = wire.raw_tensor_to_numpy(*r["x"])
+        assert y.shape == (1, 2, 8)
+        assert np.allclose(y, x + 1.0, atol=0.02)
+        assert y.dtype == np.float32 if dtype == wire.DT_F32 else True
+
+        # wrong layer set -> WorkerError (worker.rs:490-503)
+        r = await call({"type": wire.MSG_BATCH,
+                        "x": wire.numpy_to_raw(x, dtype),
+                        "batch": [("model.layers.9", 0, 9)]})
+        assert r["type"] == wire.MSG_WORKER_ERROR
+
+        # goodbye clears state (worker.rs:364-384)
+        r = await call({"type": wire.MSG_GOODBYE})
+        assert r["type"] == wire.MSG_GOODBYE
+        assert eng.resets == 1
+        writer.close()
+        server.close()
+        await server.wait_closed()
+
+    asyncio.run(asyncio.wait_for(run(), timeout=60))
+
+
+def test_bad_magic_rejected():
+    async def run():
+        worker = wire.WireWorker(EchoEngine(), ["model.layers.0"])
+        server = await asyncio.start_server(worker.handle, "127.0.0.1", 0)
+        port = server.sockets[0].getsockname()[1]
+        reader, writer = await asyncio.open_connection("127.0.0.1", port)
+        writer.write(struct.pack(">Q", (0xDEAD << 32) | 4) + b"\x00" * 4)
+        await writer.drain()
+        data = await reader.read(100)   # worker drops the connection
+        assert data == b""
+        writer.close()
+        server.close()
+        await server.wait_closed()
+
+    asyncio.run(asyncio.wait_for(run(), timeout=60))
