@@ -1021,49 +1021,56 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
     // --- phase A: scores for [sub0, sub0+TILE) --------------------------
     // 256 threads = 16 positions per pass (16 lanes per position, each
     // loading 16 B of the K row)
+    // cache-resident positions only — the newest position is handled in a
+    // dedicated step below (a per-iteration p==newest select would
+    // de-pipeline the loads; guide §5 ".s-level traps" (c))
+    const int cap = (newest >= sub0 && newest < end) ? newest : end;
 #pragma unroll 4
     for (int pass = 0; pass < TILE / 16; ++pass) {
       const int p = sub0 + pass * 16 + (t >> 4);
       float d = 0.f;
-      if (p < end) {
-        float kvv[8];
-        if (p == newest) {
-          // newest K comes RAW from the qkv buffer; norm+rope in-register
-          // (its cache slot is written below by the designated head)
+      if (p < cap && dgrp * 8 < hd) {
+        short8 kv8 = *reinterpret_cast<const short8*>(
+            kbase + (size_t)p * hd + dgrp * 8);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int dd = dgrp * 8 + j;
-            kvv[j] = dd < hd
-                         ? b2f(qkv[(size_t)nh * hd + (size_t)kvh * hd + dd])
-                         : 0.f;
-          }
-          head_norm_rope(kvv, kn, eps, crow, srow, hd, dgrp);
-          if (storer && dgrp * 8 < hd) {
-            short8 pk;
-#pragma unroll
-            for (int j = 0; j < 8; ++j) pk[j] = (short)f2b(kvv[j]);
-            *reinterpret_cast<short8*>(kbase + (size_t)newest * hd +
-                                       dgrp * 8) = pk;
-          }
-        } else if (dgrp * 8 < hd) {
-          short8 kv8 = *reinterpret_cast<const short8*>(
-              kbase + (size_t)p * hd + dgrp * 8);
-#pragma unroll
-          for (int j = 0; j < 8; ++j) kvv[j] = b2f((u16)kv8[j]);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) kvv[j] = 0.f;
-        }
-#pragma unroll
-        for (int j = 0; j < 8; ++j) d = fmaf(kvv[j], qa[j], d);
+        for (int j = 0; j < 8; ++j) d = fmaf(b2f((u16)kv8[j]), qa[j], d);
       }
       // reduce across the 16 lanes of this position
 #pragma unroll
       for (int off = 8; off >= 1; off >>= 1) d += __shfl_xor(d, off, 16);
       if (dgrp == 0)
-        stile[pass * 16 + (t >> 4)] = (p < end) ? d * scale : -INFINITY;
+        stile[pass * 16 + (t >> 4)] = (p < cap) ? d * scale : -INFINITY;
     }
     __syncthreads();
+    if (cap != end) {
+      // newest K: RAW from the qkv buffer, [qk-norm +] rope in-register;
+      // the designated head writes its cache slot for future steps
+      if (t < 16) {
+        float kvv[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int dd = t * 8 + j;
+          kvv[j] = dd < hd
+                       ? b2f(qkv[(size_t)nh * hd + (size_t)kvh * hd + dd])
+                       : 0.f;
+        }
+        head_norm_rope(kvv, kn, eps, crow, srow, hd, t);
+        if (storer && t * 8 < hd) {
+          short8 pk;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) pk[j] = (short)f2b(kvv[j]);
+          *reinterpret_cast<short8*>(kbase + (size_t)newest * hd + t * 8) =
+              pk;
+        }
+        float d = 0.f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) d = fmaf(kvv[j], qa[j], d);
+#pragma unroll
+        for (int off = 8; off >= 1; off >>= 1) d += __shfl_xor(d, off, 16);
+        if (t == 0) stile[newest - sub0] = d * scale;
+      }
+      __syncthreads();
+    }
     // --- phase B: block softmax step over the tile ----------------------
     float lm = -INFINITY;
     for (int i = t; i < TILE; i += blockDim.x) lm = fmaxf(lm, stile[i]);
@@ -1093,21 +1100,26 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
     for (int k = 0; k < TILE / 4; ++k) {
       const int po = k * 4 + wid;
       const int p = sub0 + po;
-      if (act && p < end) {
+      if (act && p < cap) {
         const float w = stile[po];
-        const u16* vr =
-            (p == newest)
-                ? qkv + (size_t)(nh + nkv) * hd + (size_t)kvh * hd + e0
-                : vbase + (size_t)p * hd + e0;
-        if (p == newest && storer) {   // store V and V^T for future steps
-          vbase[(size_t)newest * hd + e0] = vr[0];
-          vbase[(size_t)newest * hd + e0 + 1] = vr[1];
-          vtb[(size_t)e0 * max_seq + newest] = vr[0];
-          vtb[(size_t)(e0 + 1) * max_seq + newest] = vr[1];
-        }
+        const u16* vr = vbase + (size_t)p * hd + e0;
         o0 = fmaf(w, b2f(vr[0]), o0);
         o1 = fmaf(w, b2f(vr[1]), o1);
       }
+    }
+    if (cap != end && act && wid == ((newest - sub0) & 3)) {
+      // newest V: RAW from the qkv buffer; designated head stores V + V^T
+      const float w = stile[newest - sub0];
+      const u16* vr =
+          qkv + (size_t)(nh + nkv) * hd + (size_t)kvh * hd + e0;
+      if (storer) {
+        vbase[(size_t)newest * hd + e0] = vr[0];
+        vbase[(size_t)newest * hd + e0 + 1] = vr[1];
+        vtb[(size_t)e0 * max_seq + newest] = vr[0];
+        vtb[(size_t)(e0 + 1) * max_seq + newest] = vr[1];
+      }
+      o0 = fmaf(w, b2f(vr[0]), o0);
+      o1 = fmaf(w, b2f(vr[1]), o1);
     }
     __syncthreads();  // stile reused next sub-tile
   }
