@@ -339,13 +339,17 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
       launch_gemv(l.wqkv, e->x, e->qkv, nullptr, l.rms1, c.rms_eps, Nq, H, 0,
                   e->stream);
   }
-  {  // decode attention over the cache — ONE launch fusing [Qwen3 QK-norm
-     // +] rope + KV/V^T store of the newest token (no separate rope_store)
+  {  // [Qwen3 QK-norm, attention.rs:202-215, fused +] rope + KV store
+    StatScope ss(e, "rope_store", (double)Nq * hd * 0, 0);
+    launch_rope_store_decode(e->qkv, l.kc, l.vc, l.vtc, e->cos_t, e->sin_t,
+                             e->dev_pos, c.nh, c.nkv, hd, hd, e->max_seq,
+                             l.qnorm, l.knorm, c.rms_eps, e->stream);
+  }
+  {  // decode attention over the cache (single launch, split-KV combine)
     double kvbytes = 2.0 * (e->host_pos + 1) * c.skv() * 2;
     StatScope ss(e, "attn_decode", kvbytes + Sq * 2 * 2,
                  4.0 * (e->host_pos + 1) * Sq);
-    launch_attn_decode(e->qkv, l.kc, l.vc, l.vtc, e->cos_t, e->sin_t,
-                       l.qnorm, l.knorm, c.rms_eps, e->dev_pos, e->attn_ws,
+    launch_attn_decode(e->qkv, l.kc, l.vc, e->dev_pos, e->attn_ws,
                        e->attn_cnt, e->attn_out, c.nh, c.nkv, hd, e->max_seq,
                        e->nchunk, e->stream);
   }
@@ -518,8 +522,8 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
   if (layer_lo < 0 || layer_hi > c.layers || layer_lo >= layer_hi)
     return set_err(5, "bad layer range [%d,%d) of %d", layer_lo, layer_hi,
                    c.layers);
-  if (c.hd() != 16 && c.hd() != 32 && c.hd() != 64 && c.hd() != 128)
-    return set_err(5, "head_dim %d unsupported (must be 16/32/64/128)",
+  if (c.hd() > 128 || c.hd() % 8 != 0)
+    return set_err(5, "head_dim %d unsupported (must be <=128, mult of 8)",
                    c.hd());
   if (c.hidden > 16384)
     return set_err(5, "hidden_size %d unsupported (> 16384)", c.hidden);

@@ -40,16 +40,20 @@ void launch_embed_token(const u16* embed, const u32* tok, u16* x, int H,
                         hipStream_t s);
 void launch_embed_rows(const u16* embed, const u32* ids, u16* x, int S, int H,
                        hipStream_t s);
+void launch_rope_store_decode(u16* qkv, u16* kc, u16* vc, u16* vtc,
+                              const float* cost, const float* sint,
+                              const int* pos, int nh, int nkv, int hd, int rd,
+                              int max_seq, const u16* qn, const u16* kn,
+                              float eps, hipStream_t s);
 void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, u16* vtc,
                                const float* cost, const float* sint, int pos0,
                                int S, int nh, int nkv, int hd, int rd,
                                int max_seq, int qkv_stride, const u16* qn,
                                const u16* kn, float eps, hipStream_t s);
-void launch_attn_decode(u16* qkv, u16* kc, u16* vc, u16* vtc,
-                        const float* cost, const float* sint, const u16* qn,
-                        const u16* kn, float eps, const int* pos, float* ws,
-                        u32* cnt, u16* out, int nh, int nkv, int hd,
-                        int max_seq, int nchunk, hipStream_t s);
+void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
+                        const int* pos, float* ws, u32* cnt, u16* out, int nh,
+                        int nkv, int hd, int max_seq, int nchunk,
+                        hipStream_t s);
 void launch_attn_prefill(const u16* qkv, const u16* kc, const u16* vc,
                          const u16* vtc, u16* out, int S, int pos0, int nh,
                          int nkv, int hd, int max_seq, int qkv_stride,

@@ -864,6 +864,53 @@ __device__ inline float nrm(const u16* row, const u16* w, int i, float sc,
   return on ? b2f(f2b(v * sc * b2f(w[i]))) : v;
 }
 
+__global__ void k_rope_store_decode(u16* __restrict__ qkv,
+                                    u16* __restrict__ kc, u16* __restrict__ vc,
+                                    u16* __restrict__ vtc,
+                                    const float* __restrict__ cost,
+                                    const float* __restrict__ sint,
+                                    const int* __restrict__ pos, int nh,
+                                    int nkv, int hd, int rd, int max_seq,
+                                    const u16* __restrict__ qn,
+                                    const u16* __restrict__ kn, float eps) {
+  const int b = blockIdx.x;
+  const int p = *pos;
+  const int half = rd / 2;
+  const float* c = cost + (size_t)p * half;
+  const float* s = sint + (size_t)p * half;
+  if (b < nh) {                       // [qk-norm +] rope q head in place
+    u16* q = qkv + (size_t)b * hd;
+    const bool on = qn != nullptr;
+    const float sc = on ? head_norm_scale(q, hd, eps) : 1.f;
+    for (int i = threadIdx.x; i < half; i += blockDim.x) {
+      float x1 = nrm(q, qn, i, sc, on), x2 = nrm(q, qn, i + half, sc, on);
+      q[i] = f2b(x1 * c[i] - x2 * s[i]);
+      q[i + half] = f2b(x2 * c[i] + x1 * s[i]);
+    }
+  } else if (b < nh + nkv) {          // [qk-norm +] rope k head -> slot p
+    const int h = b - nh;
+    u16* k = qkv + (size_t)(nh + h) * hd;
+    u16* dst = kc + ((size_t)h * max_seq + p) * hd;
+    const bool on = kn != nullptr;
+    const float sc = on ? head_norm_scale(k, hd, eps) : 1.f;
+    for (int i = threadIdx.x; i < half; i += blockDim.x) {
+      float x1 = nrm(k, kn, i, sc, on), x2 = nrm(k, kn, i + half, sc, on);
+      dst[i] = f2b(x1 * c[i] - x2 * s[i]);
+      dst[i + half] = f2b(x2 * c[i] + x1 * s[i]);
+    }
+    for (int i = rd + threadIdx.x; i < hd; i += blockDim.x) dst[i] = k[i];
+  } else {                            // v head -> cache slot p (+ V^T)
+    const int h = b - nh - nkv;
+    const u16* v = qkv + (size_t)(nh + nkv + h) * hd;
+    u16* dst = vc + ((size_t)h * max_seq + p) * hd;
+    u16* dstt = vtc + (size_t)h * hd * max_seq + p;
+    for (int i = threadIdx.x; i < hd; i += blockDim.x) {
+      dst[i] = v[i];
+      dstt[(size_t)i * max_seq] = v[i];
+    }
+  }
+}
+
 // Prefill: S tokens at positions pos0..pos0+S-1; qkv is (S, Sq+2*Skv).
 // grid = (nh + 2*nkv, S)
 __global__ void k_rope_store_prefill(u16* __restrict__ qkv,
@@ -915,44 +962,6 @@ __global__ void k_rope_store_prefill(u16* __restrict__ qkv,
   }
 }
 
-
-// In-register [per-head rms-norm (attention.rs:202-215) +] HF half-rotation
-// rope (backends/mod.rs:470-477) for one head held by a 16-lane group:
-// lane dgrp = lane&15 owns dims dgrp*8..+8 in v[8].  All 16 lanes must be
-// active.  Requires hd in {16,32,64,128} (partner exchange mask hd/16).
-// Values are re-quantized to bf16 at each stage so the fused path is
-// bit-identical to the separate rmsnorm/rope kernels.
-__device__ inline void head_norm_rope(float (&v)[8], const u16* nw,
-                                      float eps, const float* crow,
-                                      const float* srow, int hd, int dgrp) {
-  if (nw) {
-    float ss = 0.f;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) ss += v[j] * v[j];
-#pragma unroll
-    for (int off = 8; off >= 1; off >>= 1) ss += __shfl_xor(ss, off, 16);
-    const float sc = rsqrtf(ss / (float)hd + eps);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int d = dgrp * 8 + j;
-      if (d < hd) v[j] = b2f(f2b(v[j] * sc * b2f(nw[d])));
-    }
-  }
-  const int half = hd / 2;
-  float pj[8];
-#pragma unroll
-  for (int j = 0; j < 8; ++j) pj[j] = __shfl_xor(v[j], hd >> 4, 16);
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    const int d = dgrp * 8 + j;
-    if (d < half) {
-      v[j] = b2f(f2b(v[j] * crow[d] - pj[j] * srow[d]));
-    } else if (d < hd) {
-      v[j] = b2f(f2b(v[j] * crow[d - half] + pj[j] * srow[d - half]));
-    }
-  }
-}
-
 // ---------------------------------------------------------------------------
 // Decode attention (attention.rs:300-343 semantics, GQA, f32 softmax), over
 // the preallocated cache.  Split-KV in ONE launch: grid (NCHUNK, nh); each
@@ -964,10 +973,8 @@ __device__ inline void head_norm_rope(float (&v)[8], const u16* nw,
 // K/V row loads: 64 lanes x 4 B = one coalesced 256 B transaction.
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void k_attn_decode_fused(
-    u16* __restrict__ qkv,             // (Sq|Skv|Skv), RAW (pre-rope)
-    u16* __restrict__ kc, u16* __restrict__ vc, u16* __restrict__ vtc,
-    const float* __restrict__ cost, const float* __restrict__ sint,
-    const u16* __restrict__ qn, const u16* __restrict__ kn, float eps,
+    const u16* __restrict__ q,         // (nh*hd), post-rope
+    const u16* __restrict__ kc, const u16* __restrict__ vc,
     const int* __restrict__ pos, float* __restrict__ ws,
     u32* __restrict__ cnt, u16* __restrict__ outbuf, int nh, int nkv,
     int hd, int max_seq, int nchunk) {
@@ -997,41 +1004,29 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
   __shared__ float so[4][128 + 8];
   __shared__ float stile[128];
   const float scale = rsqrtf((float)hd);
-  u16* kbase = kc + (size_t)kvh * max_seq * hd;
-  u16* vbase = vc + (size_t)kvh * max_seq * hd;
-  u16* vtb = vtc + (size_t)kvh * hd * max_seq;
-  const int newest = n - 1;            // this token's slot (pre-store)
-  const int rep = nh / nkv;
-  const bool storer = (h % rep) == 0;  // one q-head per kv-head stores
-  const float* crow = cost + (size_t)newest * (hd / 2);
-  const float* srow = sint + (size_t)newest * (hd / 2);
-  // phase-A per-thread q slice: dims (t&15)*8 .. +8 — RAW load, then fused
-  // [qk-norm +] rope (replaces the separate rope_store launch)
+  const u16* kbase = kc + (size_t)kvh * max_seq * hd;
+  const u16* vbase = vc + (size_t)kvh * max_seq * hd;
+  // phase-A per-thread q slice: dims (t&15)*8 .. +8
   const int dgrp = t & 15;
   float qa[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     const int d = dgrp * 8 + j;
-    qa[j] = d < hd ? b2f(qkv[(size_t)h * hd + d]) : 0.f;
+    qa[j] = d < hd ? b2f(q[(size_t)h * hd + d]) : 0.f;
   }
-  head_norm_rope(qa, qn, eps, crow, srow, hd, dgrp);
   float m = -INFINITY, l = 0.f, o0 = 0.f, o1 = 0.f;
   const int TILE = 128;
   for (int sub0 = start; sub0 < end; sub0 += TILE) {
     // --- phase A: scores for [sub0, sub0+TILE) --------------------------
     // 256 threads = 16 positions per pass (16 lanes per position, each
     // loading 16 B of the K row)
-    // cache-resident positions only — the newest position is handled in a
-    // dedicated step below (a per-iteration p==newest select would
-    // de-pipeline the loads; guide §5 ".s-level traps" (c))
-    const int cap = (newest >= sub0 && newest < end) ? newest : end;
 #pragma unroll 4
     for (int pass = 0; pass < TILE / 16; ++pass) {
       const int p = sub0 + pass * 16 + (t >> 4);
       float d = 0.f;
-      if (p < cap && dgrp * 8 < hd) {
-        short8 kv8 = *reinterpret_cast<const short8*>(
-            kbase + (size_t)p * hd + dgrp * 8);
+      if (p < end && dgrp * 8 < hd) {
+        const u16* kr = kbase + (size_t)p * hd + dgrp * 8;
+        short8 kv8 = *reinterpret_cast<const short8*>(kr);
 #pragma unroll
         for (int j = 0; j < 8; ++j) d = fmaf(b2f((u16)kv8[j]), qa[j], d);
       }
@@ -1039,38 +1034,9 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
 #pragma unroll
       for (int off = 8; off >= 1; off >>= 1) d += __shfl_xor(d, off, 16);
       if (dgrp == 0)
-        stile[pass * 16 + (t >> 4)] = (p < cap) ? d * scale : -INFINITY;
+        stile[pass * 16 + (t >> 4)] = (p < end) ? d * scale : -INFINITY;
     }
     __syncthreads();
-    if (cap != end) {
-      // newest K: RAW from the qkv buffer, [qk-norm +] rope in-register;
-      // the designated head writes its cache slot for future steps
-      if (t < 16) {
-        float kvv[8];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int dd = t * 8 + j;
-          kvv[j] = dd < hd
-                       ? b2f(qkv[(size_t)nh * hd + (size_t)kvh * hd + dd])
-                       : 0.f;
-        }
-        head_norm_rope(kvv, kn, eps, crow, srow, hd, t);
-        if (storer && t * 8 < hd) {
-          short8 pk;
-#pragma unroll
-          for (int j = 0; j < 8; ++j) pk[j] = (short)f2b(kvv[j]);
-          *reinterpret_cast<short8*>(kbase + (size_t)newest * hd + t * 8) =
-              pk;
-        }
-        float d = 0.f;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) d = fmaf(kvv[j], qa[j], d);
-#pragma unroll
-        for (int off = 8; off >= 1; off >>= 1) d += __shfl_xor(d, off, 16);
-        if (t == 0) stile[newest - sub0] = d * scale;
-      }
-      __syncthreads();
-    }
     // --- phase B: block softmax step over the tile ----------------------
     float lm = -INFINITY;
     for (int i = t; i < TILE; i += blockDim.x) lm = fmaxf(lm, stile[i]);
@@ -1100,26 +1066,12 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
     for (int k = 0; k < TILE / 4; ++k) {
       const int po = k * 4 + wid;
       const int p = sub0 + po;
-      if (act && p < cap) {
+      if (act && p < end) {
         const float w = stile[po];
         const u16* vr = vbase + (size_t)p * hd + e0;
         o0 = fmaf(w, b2f(vr[0]), o0);
         o1 = fmaf(w, b2f(vr[1]), o1);
       }
-    }
-    if (cap != end && act && wid == ((newest - sub0) & 3)) {
-      // newest V: RAW from the qkv buffer; designated head stores V + V^T
-      const float w = stile[newest - sub0];
-      const u16* vr =
-          qkv + (size_t)(nh + nkv) * hd + (size_t)kvh * hd + e0;
-      if (storer) {
-        vbase[(size_t)newest * hd + e0] = vr[0];
-        vbase[(size_t)newest * hd + e0 + 1] = vr[1];
-        vtb[(size_t)e0 * max_seq + newest] = vr[0];
-        vtb[(size_t)(e0 + 1) * max_seq + newest] = vr[1];
-      }
-      o0 = fmaf(w, b2f(vr[0]), o0);
-      o1 = fmaf(w, b2f(vr[1]), o1);
     }
     __syncthreads();  // stile reused next sub-tile
   }
@@ -1957,6 +1909,15 @@ void launch_embed_rows(const u16* embed, const u32* ids, u16* x, int S, int H,
                        hipStream_t s) {
   hipLaunchKernelGGL(k_embed_rows, dim3(S), dim3(256), 0, s, embed, ids, x, H);
 }
+void launch_rope_store_decode(u16* qkv, u16* kc, u16* vc, u16* vtc,
+                              const float* cost, const float* sint,
+                              const int* pos, int nh, int nkv, int hd, int rd,
+                              int max_seq, const u16* qn, const u16* kn,
+                              float eps, hipStream_t s) {
+  hipLaunchKernelGGL(k_rope_store_decode, dim3(nh + 2 * nkv), dim3(64), 0, s,
+                     qkv, kc, vc, vtc, cost, sint, pos, nh, nkv, hd, rd,
+                     max_seq, qn, kn, eps);
+}
 void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, u16* vtc,
                                const float* cost, const float* sint, int pos0,
                                int S, int nh, int nkv, int hd, int rd,
@@ -1966,14 +1927,13 @@ void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, u16* vtc,
                      s, qkv, kc, vc, vtc, cost, sint, pos0, nh, nkv, hd, rd,
                      max_seq, qkv_stride, qn, kn, eps);
 }
-void launch_attn_decode(u16* qkv, u16* kc, u16* vc, u16* vtc,
-                        const float* cost, const float* sint, const u16* qn,
-                        const u16* kn, float eps, const int* pos, float* ws,
-                        u32* cnt, u16* out, int nh, int nkv, int hd,
-                        int max_seq, int nchunk, hipStream_t s) {
+void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
+                        const int* pos, float* ws, u32* cnt, u16* out, int nh,
+                        int nkv, int hd, int max_seq, int nchunk,
+                        hipStream_t s) {
   hipLaunchKernelGGL(k_attn_decode_fused, dim3(nchunk, nh), dim3(256), 0, s,
-                     qkv, kc, vc, vtc, cost, sint, qn, kn, eps, pos, ws, cnt,
-                     out, nh, nkv, hd, max_seq, nchunk);
+                     q, kc, vc, pos, ws, cnt, out, nh, nkv, hd, max_seq,
+                     nchunk);
 }
 void launch_attn_prefill(const u16* qkv, const u16* kc, const u16* vc,
                          const u16* vtc, u16* out, int S, int pos0, int nh,
