@@ -1,0 +1,550 @@
+#include "kernels_common.h"
+#include "kernels.h"
+
+
+// ---------------------------------------------------------------------------
+// RoPE + KV store.  HF half-rotation (backends/mod.rs:470-477):
+//   out[i] = x1*c - x2*s ; out[i+half] = x2*c + x1*s, cos/sin row = position.
+// Decode: one token at *pos; K/V written into the preallocated cache at slot
+// *pos (replaces cache.rs:195-196 cat).  qkv layout: (Sq | Skv | Skv) per row.
+// grid = nh + 2*nkv blocks.
+// ---------------------------------------------------------------------------
+// Optional fused per-head QK-norm (attention.rs:202-215): applied to the
+// head row before the rotation, re-quantized to bf16 so it matches the
+// separate rmsnorm-kernel path bit-exactly.  One wave per head.
+__device__ inline float head_norm_scale(const u16* row, int hd, float eps) {
+  float ss = 0.f;
+  for (int i = threadIdx.x; i < hd; i += 64) {
+    float f = b2f(row[i]);
+    ss += f * f;
+  }
+  ss = wave_sum(ss);
+  ss = __shfl(ss, 0, WAVE);
+  return rsqrtf(ss / (float)hd + eps);
+}
+__device__ inline float nrm(const u16* row, const u16* w, int i, float sc,
+                            bool on) {
+  float v = b2f(row[i]);
+  return on ? b2f(f2b(v * sc * b2f(w[i]))) : v;
+}
+
+__global__ void k_rope_store_decode(u16* __restrict__ qkv,
+                                    u16* __restrict__ kc, u16* __restrict__ vc,
+                                    u16* __restrict__ vtc,
+                                    const float* __restrict__ cost,
+                                    const float* __restrict__ sint,
+                                    const int* __restrict__ pos, int nh,
+                                    int nkv, int hd, int rd, int max_seq,
+                                    const u16* __restrict__ qn,
+                                    const u16* __restrict__ kn, float eps) {
+  const int b = blockIdx.x;
+  const int p = *pos;
+  const int half = rd / 2;
+  const float* c = cost + (size_t)p * half;
+  const float* s = sint + (size_t)p * half;
+  if (b < nh) {                       // [qk-norm +] rope q head in place
+    u16* q = qkv + (size_t)b * hd;
+    const bool on = qn != nullptr;
+    const float sc = on ? head_norm_scale(q, hd, eps) : 1.f;
+    for (int i = threadIdx.x; i < half; i += blockDim.x) {
+      float x1 = nrm(q, qn, i, sc, on), x2 = nrm(q, qn, i + half, sc, on);
+      q[i] = f2b(x1 * c[i] - x2 * s[i]);
+      q[i + half] = f2b(x2 * c[i] + x1 * s[i]);
+    }
+  } else if (b < nh + nkv) {          // [qk-norm +] rope k head -> slot p
+    const int h = b - nh;
+    u16* k = qkv + (size_t)(nh + h) * hd;
+    u16* dst = kc + ((size_t)h * max_seq + p) * hd;
+    const bool on = kn != nullptr;
+    const float sc = on ? head_norm_scale(k, hd, eps) : 1.f;
+    for (int i = threadIdx.x; i < half; i += blockDim.x) {
+      float x1 = nrm(k, kn, i, sc, on), x2 = nrm(k, kn, i + half, sc, on);
+      dst[i] = f2b(x1 * c[i] - x2 * s[i]);
+      dst[i + half] = f2b(x2 * c[i] + x1 * s[i]);
+    }
+    for (int i = rd + threadIdx.x; i < hd; i += blockDim.x) dst[i] = k[i];
+  } else {                            // v head -> cache slot p (+ V^T)
+    const int h = b - nh - nkv;
+    const u16* v = qkv + (size_t)(nh + nkv + h) * hd;
+    u16* dst = vc + ((size_t)h * max_seq + p) * hd;
+    u16* dstt = vtc + (size_t)h * hd * max_seq + p;
+    for (int i = threadIdx.x; i < hd; i += blockDim.x) {
+      dst[i] = v[i];
+      dstt[(size_t)i * max_seq] = v[i];
+    }
+  }
+}
+
+// Prefill: S tokens at positions pos0..pos0+S-1; qkv is (S, Sq+2*Skv).
+// grid = (nh + 2*nkv, S)
+__global__ void k_rope_store_prefill(u16* __restrict__ qkv,
+                                     u16* __restrict__ kc, u16* __restrict__ vc,
+                                     u16* __restrict__ vtc,
+                                     const float* __restrict__ cost,
+                                     const float* __restrict__ sint, int pos0,
+                                     int nh, int nkv, int hd, int rd,
+                                     int max_seq, int qkv_stride,
+                                     const u16* __restrict__ qn,
+                                     const u16* __restrict__ kn, float eps) {
+  const int b = blockIdx.x;
+  const int sidx = blockIdx.y;
+  const int p = pos0 + sidx;
+  const int half = rd / 2;
+  const float* c = cost + (size_t)p * half;
+  const float* s = sint + (size_t)p * half;
+  u16* row = qkv + (size_t)sidx * qkv_stride;
+  if (b < nh) {
+    u16* q = row + (size_t)b * hd;
+    const bool on = qn != nullptr;
+    const float sc = on ? head_norm_scale(q, hd, eps) : 1.f;
+    for (int i = threadIdx.x; i < half; i += blockDim.x) {
+      float x1 = nrm(q, qn, i, sc, on), x2 = nrm(q, qn, i + half, sc, on);
+      q[i] = f2b(x1 * c[i] - x2 * s[i]);
+      q[i + half] = f2b(x2 * c[i] + x1 * s[i]);
+    }
+  } else if (b < nh + nkv) {
+    const int h = b - nh;
+    u16* k = row + (size_t)(nh + h) * hd;
+    u16* dst = kc + ((size_t)h * max_seq + p) * hd;
+    const bool on = kn != nullptr;
+    const float sc = on ? head_norm_scale(k, hd, eps) : 1.f;
+    for (int i = threadIdx.x; i < half; i += blockDim.x) {
+      float x1 = nrm(k, kn, i, sc, on), x2 = nrm(k, kn, i + half, sc, on);
+      dst[i] = f2b(x1 * c[i] - x2 * s[i]);
+      dst[i + half] = f2b(x2 * c[i] + x1 * s[i]);
+    }
+    for (int i = rd + threadIdx.x; i < hd; i += blockDim.x) dst[i] = k[i];
+  } else {
+    const int h = b - nh - nkv;
+    const u16* v = row + (size_t)(nh + nkv + h) * hd;
+    u16* dst = vc + ((size_t)h * max_seq + p) * hd;
+    u16* dstt = vtc + (size_t)h * hd * max_seq + p;
+    for (int i = threadIdx.x; i < hd; i += blockDim.x) {
+      dst[i] = v[i];
+      dstt[(size_t)i * max_seq] = v[i];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Decode attention (attention.rs:300-343 semantics, GQA, f32 softmax), over
+// the preallocated cache.  Split-KV in ONE launch: grid (NCHUNK, nh); each
+// block computes an online-softmax partial over a contiguous slice of
+// positions into ws[h][chunk] = {o[hd], m, l}; the LAST-arriving block of a
+// head combines the partials (agent-scope release/acquire + arrival counter
+// per cdna_hip_programming.md §6 Guideline 16 — placement-independent).
+// cnt[nh] must be zeroed before every launch (hipMemsetAsync node).
+// K/V row loads: 64 lanes x 4 B = one coalesced 256 B transaction.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_attn_decode_fused(
+    const u16* __restrict__ q,         // (nh*hd), post-rope
+    const u16* __restrict__ kc, const u16* __restrict__ vc,
+    const int* __restrict__ pos, float* __restrict__ ws,
+    u32* __restrict__ cnt, u16* __restrict__ outbuf, int nh, int nkv,
+    int hd, int max_seq, int nchunk) {
+  const int h = blockIdx.y;
+  const int chunk = blockIdx.x;
+  const int n = *pos + 1;
+  const int cs = (n + nchunk - 1) / nchunk;
+  const int start = chunk * cs;
+  const int end = min(start + cs, n);
+  const int kvh = h / (nh / nkv);
+  const int t = threadIdx.x, wid = t / WAVE, lane = t % WAVE;
+  const int e0 = 2 * lane;             // dims (2*lane, 2*lane+1)
+  const bool act = e0 + 1 < hd;
+  float* wsrow = ws + ((size_t)h * nchunk + chunk) * (hd + 2);
+#define WS_STORE(p, v)                                                     \
+  __hip_atomic_store((p), (v), __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)
+#define WS_LOAD(p)                                                         \
+  __hip_atomic_load((p), __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)
+
+  // Tiled two-phase structure: the serial online-softmax chain runs once
+  // per TILE (128 positions), not per position — phase A computes a tile's
+  // scores with 16-B coalesced K loads (16 lanes per position), phase B is
+  // a block-wide softmax step, phase C accumulates PV with coalesced,
+  // mutually independent V loads.  The per-position serial chain of the
+  // naive version measured 128 GB/s at n=2048.
+  __shared__ float sm[4], sl[4];
+  __shared__ float so[4][128 + 8];
+  __shared__ float stile[128];
+  const float scale = rsqrtf((float)hd);
+  const u16* kbase = kc + (size_t)kvh * max_seq * hd;
+  const u16* vbase = vc + (size_t)kvh * max_seq * hd;
+  // phase-A per-thread q slice: dims (t&15)*8 .. +8
+  const int dgrp = t & 15;
+  float qa[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int d = dgrp * 8 + j;
+    qa[j] = d < hd ? b2f(q[(size_t)h * hd + d]) : 0.f;
+  }
+  float m = -INFINITY, l = 0.f, o0 = 0.f, o1 = 0.f;
+  const int TILE = 128;
+  for (int sub0 = start; sub0 < end; sub0 += TILE) {
+    // --- phase A: scores for [sub0, sub0+TILE) --------------------------
+    // 256 threads = 16 positions per pass (16 lanes per position, each
+    // loading 16 B of the K row)
+#pragma unroll 4
+    for (int pass = 0; pass < TILE / 16; ++pass) {
+      const int p = sub0 + pass * 16 + (t >> 4);
+      float d = 0.f;
+      if (p < end && dgrp * 8 < hd) {
+        const u16* kr = kbase + (size_t)p * hd + dgrp * 8;
+        short8 kv8 = *reinterpret_cast<const short8*>(kr);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) d = fmaf(b2f((u16)kv8[j]), qa[j], d);
+      }
+      // reduce across the 16 lanes of this position
+#pragma unroll
+      for (int off = 8; off >= 1; off >>= 1) d += __shfl_xor(d, off, 16);
+      if (dgrp == 0)
+        stile[pass * 16 + (t >> 4)] = (p < end) ? d * scale : -INFINITY;
+    }
+    __syncthreads();
+    // --- phase B: block softmax step over the tile ----------------------
+    float lm = -INFINITY;
+    for (int i = t; i < TILE; i += blockDim.x) lm = fmaxf(lm, stile[i]);
+    lm = wave_max(lm);
+    if (lane == 0) sm[wid] = lm;
+    __syncthreads();
+    const float tmax = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
+    const float mnew = fmaxf(m, tmax);
+    const float alpha = (mnew == -INFINITY) ? 0.f : __expf(m - mnew);
+    float psum = 0.f;
+    for (int i = t; i < TILE; i += blockDim.x) {
+      const float sv = stile[i];
+      const float e = (sv == -INFINITY) ? 0.f : __expf(sv - mnew);
+      stile[i] = e;
+      psum += e;
+    }
+    psum = wave_sum(psum);
+    if (lane == 0) sl[wid] = psum;
+    __syncthreads();
+    l = l * alpha + sl[0] + sl[1] + sl[2] + sl[3];
+    o0 *= alpha;
+    o1 *= alpha;
+    m = mnew;
+    // --- phase C: PV accumulate (thread t: dims 2*lane, position residue
+    // wid mod 4; loads independent across iterations) --------------------
+#pragma unroll 8
+    for (int k = 0; k < TILE / 4; ++k) {
+      const int po = k * 4 + wid;
+      const int p = sub0 + po;
+      if (act && p < end) {
+        const float w = stile[po];
+        const u16* vr = vbase + (size_t)p * hd + e0;
+        o0 = fmaf(w, b2f(vr[0]), o0);
+        o1 = fmaf(w, b2f(vr[1]), o1);
+      }
+    }
+    __syncthreads();  // stile reused next sub-tile
+  }
+  // combine the 4 position-residue partials per dim
+  if (act) { so[wid][e0] = o0; so[wid][e0 + 1] = o1; }
+  __syncthreads();
+  if (t == 0) {
+    WS_STORE(&wsrow[hd], m);
+    WS_STORE(&wsrow[hd + 1], l);
+  }
+  for (int d = t; d < hd; d += blockDim.x)
+    WS_STORE(&wsrow[d], so[0][d] + so[1][d] + so[2][d] + so[3][d]);
+
+  // ---- publish partial + elect the combining block ----------------------
+  // sc1 write-through publish (Guideline 16 R1 variant): the ws stores above
+  // are agent-scope relaxed 4-B atomics (= sc1 stores, the natural width
+  // here), so no release fence is needed — just drain, then count arrivals.
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // every wave drains
+  __syncthreads();
+  if (t == 0) {
+    // epoch-free election: counters monotonically accumulate, the block
+    // drawing (v % nchunk) == nchunk-1 combines — no per-launch reset
+    u32 v = __hip_atomic_fetch_add(&cnt[h], 1u, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+    sm[0] = (v % (u32)nchunk == (u32)(nchunk - 1)) ? 1.f : 0.f;
+  }
+  __syncthreads();
+  if (sm[0] == 0.f) return;
+  // reducer reads the slabs with sc1 loads — no acquire fence, no L1 risk
+
+  // ---- combine this head's partials (runs in exactly one block) ----------
+  float* base = ws + (size_t)h * nchunk * (hd + 2);
+  // stage m,l in LDS (parallel sc1 loads; serial dependent uncached loads
+  // were the reducer's cost), then combine with per-chunk weights from LDS
+  if (t < nchunk) {
+    sm[0] = 0.f;  // keep sm[0] clear; use so rows as staging
+    so[0][t] = WS_LOAD(&base[t * (hd + 2) + hd]);
+    so[1][t] = WS_LOAD(&base[t * (hd + 2) + hd + 1]);
+  }
+  __syncthreads();
+  float M = -INFINITY;
+  for (int c = 0; c < nchunk; ++c) M = fmaxf(M, so[0][c]);
+  float L = 0.f;
+  for (int c = 0; c < nchunk; ++c)
+    if (so[0][c] != -INFINITY) L += so[1][c] * __expf(so[0][c] - M);
+  for (int d = t; d < hd; d += blockDim.x) {
+    float o = 0.f;
+#pragma unroll 4
+    for (int c = 0; c < nchunk; ++c)
+      if (so[0][c] != -INFINITY)
+        o += WS_LOAD(&base[c * (hd + 2) + d]) * __expf(so[0][c] - M);
+    outbuf[(size_t)h * hd + d] = f2b(o / L);
+  }
+#undef WS_STORE
+#undef WS_LOAD
+}
+
+// ---------------------------------------------------------------------------
+// Prefill attention — flash-style f32 online softmax, causal, GQA, over the
+// cache (which already holds positions [0, pos0+S)).  One wave per query
+// row; 4 rows per block.  (attention.rs:300-343 + cache.rs:150-160 mask.)
+// q rows come from the post-rope qkv buffer.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_attn_prefill(
+    const u16* __restrict__ qkv, const u16* __restrict__ kc,
+    const u16* __restrict__ vc, u16* __restrict__ out, int S, int pos0,
+    int nh, int nkv, int hd, int max_seq, int qkv_stride, int out_stride) {
+  const int h = blockIdx.y;
+  const int t = threadIdx.x, wid = t / WAVE, lane = t % WAVE;
+  const int sidx = blockIdx.x * 4 + wid;
+  if (sidx >= S) return;
+  const int kvh = h / (nh / nkv);
+  const int e0 = 2 * lane;
+  const bool act = e0 + 1 < hd;
+  const int n = pos0 + sidx + 1;      // causal: attend to <= own position
+  float q0 = 0.f, q1 = 0.f;
+  if (act) {
+    const u16* qr = qkv + (size_t)sidx * qkv_stride + (size_t)h * hd + e0;
+    q0 = b2f(qr[0]);
+    q1 = b2f(qr[1]);
+  }
+  const float scale = rsqrtf((float)hd);
+  float m = -INFINITY, l = 0.f, o0 = 0.f, o1 = 0.f;
+  const u16* kbase = kc + (size_t)kvh * max_seq * hd;
+  const u16* vbase = vc + (size_t)kvh * max_seq * hd;
+  for (int p = 0; p < n; ++p) {
+    float dot = 0.f;
+    if (act) {
+      const u16* kr = kbase + (size_t)p * hd + e0;
+      dot = q0 * b2f(kr[0]) + q1 * b2f(kr[1]);
+    }
+    dot = wave_sum(dot) * scale;
+    dot = __shfl(dot, 0, WAVE);
+    float mn = fmaxf(m, dot);
+    float alpha = __expf(m - mn);
+    float pw = __expf(dot - mn);
+    float v0 = 0.f, v1 = 0.f;
+    if (act) {
+      const u16* vr = vbase + (size_t)p * hd + e0;
+      v0 = b2f(vr[0]);
+      v1 = b2f(vr[1]);
+    }
+    o0 = o0 * alpha + pw * v0;
+    o1 = o1 * alpha + pw * v1;
+    l = l * alpha + pw;
+    m = mn;
+  }
+  if (act) {
+    u16* orow = out + (size_t)sidx * out_stride + (size_t)h * hd + e0;
+    orow[0] = f2b(o0 / l);
+    orow[1] = f2b(o1 / l);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// MFMA flash-attention prefill (hd == 128) — causal, GQA, f32 online
+// softmax, bf16 I/O.  Structure after the guide's 8-wave 32x32 ladder
+// (cdna_hip_programming.md §B "Fused attention prefill"), basic variant:
+//   - 8 waves per workgroup, each wave owns 32 query rows of one head
+//   - swapped QK^T: P = mfma(K_tile, Q_tile) so the softmax column is
+//     lane-local (col j = q = lane&31); 32x32x16 bf16 MFMA, 8 per tile
+//   - P -> bf16 pack + __shfl_xor(32) half-exchange assembles the PV
+//     A-fragments in-register (the T12 idea without inline asm)
+//   - PV reads V from a TRANSPOSED cache (vt[kvh][d][pos]) so the
+//     B-fragment's 8-consecutive-k elements are one 16-B load
+// Per 32-kv tile per wave: 16 MFMAs + ~16 16-B global loads; no LDS, no
+// barriers (waves fully independent).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(512) void k_attn_prefill_mfma(
+    const u16* __restrict__ qkv, const u16* __restrict__ kc,
+    const u16* __restrict__ vtc, u16* __restrict__ out, int S, int pos0,
+    int nh, int nkv, int max_seq, int qkv_stride, int out_stride) {
+  const int hd = 128;
+  const int w = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int lhalf = lane >> 5, lq = lane & 31;
+  const int h = blockIdx.y;
+  const int kvh = h / (nh / nkv);
+  const int qb = blockIdx.x * 256 + w * 32;
+  if (qb >= S) return;
+
+  // Q fragments (persistent): B[k][j=q], lane holds q = lq, dims
+  // kk*16 + lhalf*8 .. +8
+  bf16x8 qf[8];
+  {
+    const int row = min(qb + lq, S - 1);
+    const u16* qrow = qkv + (size_t)row * qkv_stride + (size_t)h * hd;
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk)
+      qf[kk] = *reinterpret_cast<const bf16x8*>(qrow + kk * 16 + lhalf * 8);
+  }
+
+  f32x16 oacc[4];
+#pragma unroll
+  for (int db = 0; db < 4; ++db)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) oacc[db][r] = 0.f;
+  float m = -INFINITY, l = 0.f;
+  const float scale = rsqrtf((float)hd);
+  const int q_abs = pos0 + qb + lq;
+  const bool q_valid = qb + lq < S;
+  const int n_wave = pos0 + min(qb + 32, S);  // kv needed by this wave
+  const int ntiles = (n_wave + 31) / 32;
+  const u16* kbase = kc + (size_t)kvh * max_seq * hd;
+  const u16* vtbase = vtc + (size_t)kvh * hd * max_seq;
+
+  for (int tile = 0; tile < ntiles; ++tile) {
+    const int pkv = tile * 32;
+    // K fragments: A[i=kv][k], lane holds kv = lq
+    bf16x8 kf[8];
+    {
+      const u16* krow = kbase + (size_t)(pkv + lq) * hd;
+#pragma unroll
+      for (int kk = 0; kk < 8; ++kk)
+        kf[kk] =
+            *reinterpret_cast<const bf16x8*>(krow + kk * 16 + lhalf * 8);
+    }
+    f32x16 p;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) p[r] = 0.f;
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk)
+      p = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[kk], qf[kk], p, 0, 0, 0);
+    // scale + causal mask (reg r -> kv row (r&3)+8*(r>>2)+4*lhalf)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kv_abs = pkv + (r & 3) + 8 * (r >> 2) + 4 * lhalf;
+      p[r] = (q_valid && kv_abs <= q_abs) ? p[r] * scale : -INFINITY;
+    }
+    // online softmax stats for column q = lq (halves combined via xor-32)
+    float tm = -INFINITY;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) tm = fmaxf(tm, p[r]);
+    tm = fmaxf(tm, __shfl_xor(tm, 32, WAVE));
+    const float mnew = fmaxf(m, tm);
+    const float alpha = (mnew == -INFINITY) ? 0.f : __expf(m - mnew);
+    float ep[16];
+    float tsum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      ep[r] = (p[r] == -INFINITY) ? 0.f : __expf(p[r] - mnew);
+      tsum += ep[r];
+    }
+    tsum += __shfl_xor(tsum, 32, WAVE);
+    l = l * alpha + tsum;
+    m = mnew;
+    // pack expP to bf16 pairs and exchange halves: after this each lane can
+    // assemble A[i=q=lq][k=kv] fragments for PV
+    u32 pk[8], rcv[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      pk[i] = (u32)f2b(ep[2 * i]) | ((u32)f2b(ep[2 * i + 1]) << 16);
+      rcv[i] = __shfl_xor(pk[i], 32, WAVE);
+    }
+    // per-row alpha for the O rescale (row q via lane shuffle)
+    float arow[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      arow[r] = __shfl(alpha, (r & 3) + 8 * (r >> 2) + 4 * lhalf, WAVE);
+#pragma unroll
+    for (int db = 0; db < 4; ++db)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) oacc[db][r] *= arow[r];
+    // PV: two K=16 windows over the 32-kv tile
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      union { u32 u[4]; bf16x8 v; } af;
+      if (lhalf == 0) {
+        af.u[0] = pk[4 * kk];
+        af.u[1] = pk[4 * kk + 1];
+        af.u[2] = rcv[4 * kk];
+        af.u[3] = rcv[4 * kk + 1];
+      } else {
+        af.u[0] = rcv[4 * kk + 2];
+        af.u[1] = rcv[4 * kk + 3];
+        af.u[2] = pk[4 * kk + 2];
+        af.u[3] = pk[4 * kk + 3];
+      }
+#pragma unroll
+      for (int db = 0; db < 4; ++db) {
+        const int d = db * 32 + lq;
+        const u16* vt =
+            vtbase + (size_t)d * max_seq + pkv + 16 * kk + lhalf * 8;
+        bf16x8 vf = *reinterpret_cast<const bf16x8*>(vt);
+        oacc[db] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(af.v, vf, oacc[db], 0,
+                                                    0, 0);
+      }
+    }
+  }
+
+  // epilogue: divide by l (per q row, via shuffle) and store
+  float lrow[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    float lv = __shfl(l, (r & 3) + 8 * (r >> 2) + 4 * lhalf, WAVE);
+    lrow[r] = 1.f / lv;
+  }
+#pragma unroll
+  for (int db = 0; db < 4; ++db) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qrow = (r & 3) + 8 * (r >> 2) + 4 * lhalf;
+      const int srow = qb + qrow;
+      if (srow < S)
+        out[(size_t)srow * out_stride + (size_t)h * hd + db * 32 + lq] =
+            f2b(oacc[db][r] * lrow[r]);
+    }
+  }
+}
+
+void launch_rope_store_decode(u16* qkv, u16* kc, u16* vc, u16* vtc,
+                              const float* cost, const float* sint,
+                              const int* pos, int nh, int nkv, int hd, int rd,
+                              int max_seq, const u16* qn, const u16* kn,
+                              float eps, hipStream_t s) {
+  hipLaunchKernelGGL(k_rope_store_decode, dim3(nh + 2 * nkv), dim3(64), 0, s,
+                     qkv, kc, vc, vtc, cost, sint, pos, nh, nkv, hd, rd,
+                     max_seq, qn, kn, eps);
+}
+void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, u16* vtc,
+                               const float* cost, const float* sint, int pos0,
+                               int S, int nh, int nkv, int hd, int rd,
+                               int max_seq, int qkv_stride, const u16* qn,
+                               const u16* kn, float eps, hipStream_t s) {
+  hipLaunchKernelGGL(k_rope_store_prefill, dim3(nh + 2 * nkv, S), dim3(64), 0,
+                     s, qkv, kc, vc, vtc, cost, sint, pos0, nh, nkv, hd, rd,
+                     max_seq, qkv_stride, qn, kn, eps);
+}
+void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
+                        const int* pos, float* ws, u32* cnt, u16* out, int nh,
+                        int nkv, int hd, int max_seq, int nchunk,
+                        hipStream_t s) {
+  hipLaunchKernelGGL(k_attn_decode_fused, dim3(nchunk, nh), dim3(256), 0, s,
+                     q, kc, vc, pos, ws, cnt, out, nh, nkv, hd, max_seq,
+                     nchunk);
+}
+void launch_attn_prefill(const u16* qkv, const u16* kc, const u16* vc,
+                         const u16* vtc, u16* out, int S, int pos0, int nh,
+                         int nkv, int hd, int max_seq, int qkv_stride,
+                         int out_stride, hipStream_t s) {
+  if (hd == 128) {
+    hipLaunchKernelGGL(k_attn_prefill_mfma, dim3((S + 255) / 256, nh),
+                       dim3(512), 0, s, qkv, kc, vtc, out, S, pos0, nh, nkv,
+                       max_seq, qkv_stride, out_stride);
+  } else {
+    hipLaunchKernelGGL(k_attn_prefill, dim3((S + 3) / 4, nh), dim3(256), 0, s,
+                       qkv, kc, vc, out, S, pos0, nh, nkv, hd, max_seq,
+                       qkv_stride, out_stride);
+  }
+}

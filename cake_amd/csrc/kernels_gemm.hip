@@ -1,0 +1,310 @@
+#include "kernels_common.h"
+#include "kernels.h"
+
+
+// ---------------------------------------------------------------------------
+// Prefill GEMM — MFMA bf16.  C[M,N] = A[M,K] @ W[N,K]^T (+ residual).
+// 128x128 tile, BK=64, 256 threads (4 waves as 2x2 of 64x64 sub-tiles),
+// v_mfma_f32_16x16x32_bf16, double-buffered LDS filled by
+// global_load_lds_dwordx4 with the XOR source-swizzle (rule 21) so the
+// fragment ds_read_b128 is bank-conflict-free (guide §5 ladder step 3).
+// Row clamp handles M/N tails; requires K % 64 == 0 (all hot-path K are).
+// ---------------------------------------------------------------------------
+#define GEMM_BM 128
+#define GEMM_BN 128
+#define GEMM_BK 64
+
+template <int EPI>  // 0: bf16 store, 1: bf16 store + residual add
+__global__ __launch_bounds__(256) void k_gemm_bf16(
+    const u16* __restrict__ A, const u16* __restrict__ W, u16* __restrict__ C,
+    const u16* __restrict__ res, int M, int N, int K) {
+  // dynamic-LDS base must be 16-B aligned (guide G17)
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  u16* As = reinterpret_cast<u16*>(smem);                       // [2][128][64]
+  u16* Bs = reinterpret_cast<u16*>(smem + 2 * GEMM_BM * GEMM_BK * 2);
+
+  // XCD-aware block swizzle (guide T1, bijective variant): contiguous
+  // chunks of the grid per XCD so neighbor tiles hit the same per-XCD L2
+  int mt, nt;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    const int orig = blockIdx.x + gridDim.x * blockIdx.y;
+    const int q = nwg / 8, rr = nwg % 8;
+    const int xcd = orig % 8, idx = orig / 8;
+    const int wgid =
+        (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+    mt = wgid % gridDim.x;
+    nt = wgid / gridDim.x;
+  }
+  const int m0 = mt * GEMM_BM;
+  const int n0 = nt * GEMM_BN;
+  const int t = threadIdx.x;
+  const int wid = t / WAVE, lane = t % WAVE;
+  const int wr = wid / 2, wc = wid % 2;  // wave's 64x64 quadrant
+
+  // staging geometry: each wave issues 4 glds per 16 KB tile; instruction j
+  // of wave w writes LDS rows [(w*4+j)*8, +8) (8 rows x 128 B), lane l ->
+  // row sub = l/8, 16-B unit u = l%8, with source column unit u ^ (row & 7).
+  auto stage = [&](int buf, int kt) {
+    const int ktbase = kt * GEMM_BK;
+    u16* as = As + (size_t)buf * GEMM_BM * GEMM_BK;
+    u16* bs = Bs + (size_t)buf * GEMM_BN * GEMM_BK;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int r = (wid * 4 + j) * 8 + lane / 8;
+      const int u = (lane % 8) ^ (r & 7);
+      {  // A tile
+        int grow = min(m0 + r, M - 1);
+        const u16* src = A + (size_t)grow * K + ktbase + u * 8;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)(uintptr_t)src,
+            (__attribute__((address_space(3))) void*)(uintptr_t)(
+                as + (size_t)(wid * 4 + j) * 8 * GEMM_BK),
+            16, 0, 0);
+      }
+      {  // W tile (B^T)
+        int grow = min(n0 + r, N - 1);
+        const u16* src = W + (size_t)grow * K + ktbase + u * 8;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)(uintptr_t)src,
+            (__attribute__((address_space(3))) void*)(uintptr_t)(
+                bs + (size_t)(wid * 4 + j) * 8 * GEMM_BK),
+            16, 0, 0);
+      }
+    }
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ntiles = K / GEMM_BK;
+  stage(0, 0);
+  __syncthreads();  // drains the glds (vmcnt(0) implied by the barrier)
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < ntiles) stage(cur ^ 1, kt + 1);
+    const u16* as = As + (size_t)cur * GEMM_BM * GEMM_BK;
+    const u16* bs = Bs + (size_t)cur * GEMM_BN * GEMM_BK;
+#pragma unroll
+    for (int ks = 0; ks < GEMM_BK / 32; ++ks) {
+      bf16x8 af[4], bf[4];
+      const int kk = ks * 32 + (lane / 16) * 8;
+      const int ku = kk / 8;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int ar = wr * 64 + i * 16 + (lane % 16);
+        af[i] = *reinterpret_cast<const bf16x8*>(
+            as + (size_t)ar * GEMM_BK + (size_t)(ku ^ (ar & 7)) * 8);
+        const int br = wc * 64 + i * 16 + (lane % 16);
+        bf[i] = *reinterpret_cast<const bf16x8*>(
+            bs + (size_t)br * GEMM_BK + (size_t)(ku ^ (br & 7)) * 8);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C/D map (guide §3): col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wr * 64 + i * 16 + (lane / 16) * 4 + r;
+        const int col = n0 + wc * 64 + j * 16 + (lane % 16);
+        if (row < M && col < N) {
+          float v = acc[i][j][r];
+          if (EPI == 1) v += b2f(res[(size_t)row * N + col]);
+          C[(size_t)row * N + col] = f2b(v);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// 256x256 GEMM with counted-vmcnt pipelining (guide §5 T3/T4: the counted
+// wait IS the gain — never drain vmcnt(0) in the main loop).
+//
+// Geometry: BM=BN=256, BK=64, 8 waves (2M x 4N), per-wave 128x64 output
+// (8x4 16x16 frags, 128 acc VGPRs).  LDS 128 KB: 2 tile-buffers x
+// {A,B} x 2 row-halves x [128][64] bf16, my known-good XOR-unit swizzle.
+//
+// Schedule (uniform, every phase): stage 2 row-half pieces of tile t+1
+// (4 glds/wave) -> asm s_waitcnt vmcnt(4) (retires everything except the 4
+// just-issued, so tile t is fully resident) -> raw s_barrier -> 12
+// ds_read_b128 + 32 MFMA (one k-step) -> next phase.  Piece issue order per
+// wave: (t,A-h0)(t,A-h1) in phase ks=0, (t,B-h0)(t,B-h1) in ks=1, so the
+// FIFO count proof holds; buffers alternate by tile parity and a piece is
+// overwritten two tiles after its last read (>= 2 barriers apart).
+// Requires K % 64 == 0; M/N tails handled by clamped loads + guarded
+// stores.  Dispatched only when the grid has >= GEMM256_MIN_BLOCKS tiles.
+// ---------------------------------------------------------------------------
+template <int EPI>
+__global__ __launch_bounds__(512) void k_gemm_256(
+    const u16* __restrict__ A, const u16* __restrict__ W, u16* __restrict__ C,
+    const u16* __restrict__ res, int M, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // piece (op, buf, half): As op=0 at 0, Bs op=1 at 64 KB
+  auto piece = [&](int op, int buf, int half) -> u16* {
+    return reinterpret_cast<u16*>(smem + op * 65536 +
+                                  (buf * 2 + half) * 16384);
+  };
+  // XCD-aware bijective block swizzle (T1)
+  int mt, nt;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    const int orig = blockIdx.x + gridDim.x * blockIdx.y;
+    const int q = nwg / 8, rr = nwg % 8;
+    const int xcd = orig % 8, idx = orig / 8;
+    const int wgid =
+        (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+    mt = wgid % gridDim.x;
+    nt = wgid / gridDim.x;
+  }
+  const int m0 = mt * 256, n0 = nt * 256;
+  const int t = threadIdx.x;
+  const int w = t / WAVE, lane = t % WAVE;
+  const int wm = w >> 2, wn = w & 3;
+
+  // stage the two row-half pieces of ONE operand for tile tau
+  auto stage_op = [&](int op, int tau) {
+    const u16* G = op == 0 ? A : W;
+    const int rows = op == 0 ? M : N;
+    const int base = op == 0 ? m0 : n0;
+    const int buf = tau & 1;
+    const int kt = tau * 64;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      u16* dst = piece(op, buf, half);
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const int rloc = (w * 2 + j) * 8 + lane / 8;   // 0..127
+        const int grow = min(base + half * 128 + rloc, rows - 1);
+        const int u = (lane % 8) ^ (rloc & 7);
+        const u16* src = G + (size_t)grow * K + kt + u * 8;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)(uintptr_t)src,
+            (__attribute__((address_space(3))) void*)(uintptr_t)(
+                dst + (size_t)(w * 2 + j) * 8 * 64),
+            16, 0, 0);
+      }
+    }
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int NT = K / 64;
+  // prologue: tile 0, both operands (8 glds/wave in flight)
+  stage_op(0, 0);
+  stage_op(1, 0);
+  for (int tau = 0; tau < NT; ++tau) {
+    const int buf = tau & 1;
+    const u16* ap = piece(0, buf, wm);               // this wave's A half
+    const u16* bp = piece(1, buf, wn >> 1);          // this wave's B half
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      if (tau + 1 < NT) {
+        stage_op(ks, tau + 1);                       // ks0: A, ks1: B
+        // tile tau fully resident once all but the 4 just-issued retire
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // last tile
+      }
+      __builtin_amdgcn_s_barrier();
+      bf16x8 af[8], bf[4];
+      const int u = ks * 4 + (lane / 16);            // 16-B unit 0..7
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi) {
+        const int r = mi * 16 + (lane % 16);
+        af[mi] = *reinterpret_cast<const bf16x8*>(
+            ap + (size_t)r * 64 + (size_t)(u ^ (r & 7)) * 8);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int r = (wn & 1) * 64 + ni * 16 + (lane % 16);
+        bf[ni] = *reinterpret_cast<const bf16x8*>(
+            bp + (size_t)r * 64 + (size_t)(u ^ (r & 7)) * 8);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // quiesce before exit
+
+  // epilogue (C/D map: col = lane&15, row = (lane>>4)*4 + reg)
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wm * 128 + mi * 16 + (lane / 16) * 4 + r;
+        const int col = n0 + wn * 64 + ni * 16 + (lane % 16);
+        if (row < M && col < N) {
+          float v = acc[mi][ni][r];
+          if (EPI == 1) v += b2f(res[(size_t)row * N + col]);
+          C[(size_t)row * N + col] = f2b(v);
+        }
+      }
+    }
+  }
+}
+
+void launch_gemm(const u16* A, const u16* W, u16* C, const u16* res, int M,
+                 int N, int K, int epi, hipStream_t s) {
+  // 256^2 counted-vmcnt kernel when the grid still fills the chip at one
+  // 8-wave block per CU (CAKE_GEMM256=0 disables for A/B)
+  static const int use256 = [] {
+    const char* v = getenv("CAKE_GEMM256");
+    return v ? atoi(v) : 1;
+  }();
+  const int mt = (M + 255) / 256, nt = (N + 255) / 256;
+  if (use256 && (long)mt * nt >= 200) {
+    static bool attr_set = false;
+    if (!attr_set) {
+      hipFuncSetAttribute((const void*)&k_gemm_256<0>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+      hipFuncSetAttribute((const void*)&k_gemm_256<1>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+      attr_set = true;
+    }
+    dim3 grid(mt, nt);
+    if (epi == 0)
+      hipLaunchKernelGGL(k_gemm_256<0>, grid, dim3(512), 131072, s, A, W, C,
+                         res, M, N, K);
+    else
+      hipLaunchKernelGGL(k_gemm_256<1>, grid, dim3(512), 131072, s, A, W, C,
+                         res, M, N, K);
+    return;
+  }
+  dim3 grid((M + GEMM_BM - 1) / GEMM_BM, (N + GEMM_BN - 1) / GEMM_BN);
+  size_t lds = 2 * (size_t)(GEMM_BM + GEMM_BN) * GEMM_BK * 2;
+  if (epi == 0)
+    hipLaunchKernelGGL(k_gemm_bf16<0>, grid, dim3(256), lds, s, A, W, C, res,
+                       M, N, K);
+  else
+    hipLaunchKernelGGL(k_gemm_bf16<1>, grid, dim3(256), lds, s, A, W, C, res,
+                       M, N, K);
+}

@@ -1,0 +1,718 @@
+#include "kernels_common.h"
+#include "kernels.h"
+
+
+// ---------------------------------------------------------------------------
+// GEMV family — decode's workhorse (HBM-bound weight streaming at 16 B/lane,
+// guide §5 "GEMV / M <= 16": weights straight to VGPRs, no LDS round trip).
+//
+// k_gemv_reg<ROWS, EPI, NORM, KB>: x is loaded ONCE into registers (up to
+// KB*8 f32/thread, K <= KB*2048) and reused for every row; with NORM the
+// kernel fuses the preceding rms_norm (backends/mod.rs:244-246) into the
+// x load — the normed value is re-quantized to bf16 so the fused path is
+// bit-identical to rmsnorm-then-gemv.  ROWS per block is chosen by the
+// launcher so the grid has >= ~2048 workgroups (256 CUs want many blocks).
+// EPI: 0 = bf16 out; 1 = bf16 out + residual add; 2 = f32 out (logits).
+// ---------------------------------------------------------------------------
+template <int ROWS, int EPI, bool NORM, int KB>
+__global__ __launch_bounds__(256) void k_gemv_reg(
+    const u16* __restrict__ W, const u16* __restrict__ x,
+    void* __restrict__ out, const u16* __restrict__ res,
+    const u16* __restrict__ nw, float eps, int N, int K) {
+  const int t = threadIdx.x;
+  const int row0 = blockIdx.x * ROWS;
+  const int wid = t / WAVE, lane = t % WAVE;
+  __shared__ float red[ROWS > 4 ? ROWS : 4][4];
+
+  // For short K, issue ALL weight loads first — they stay in flight across
+  // the x/norm phase (plain VGPR loads survive s_barrier; guide §5
+  // pipelining note), hiding the norm reduction latency entirely.
+  short8 wpre[KB <= 2 ? ROWS : 1][KB <= 2 ? KB : 1];
+  if (KB <= 2) {
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+#pragma unroll
+      for (int i = 0; i < KB; ++i) {
+        const int k0 = i * 2048 + t * 8;
+        if (row0 + r < N && k0 < K)
+          wpre[r][i] = ntload8(W + (size_t)(row0 + r) * K + k0);
+      }
+  }
+
+  // phase 1: x -> registers (f32), optionally fused rms_norm
+  float xr[KB * 8];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    if (k0 < K) {
+      short8 xv = *reinterpret_cast<const short8*>(x + k0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xv[j]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = 0.f;
+    }
+  }
+  if (NORM) {
+    float ss = 0.f;
+#pragma unroll
+    for (int i = 0; i < KB * 8; ++i) ss += xr[i] * xr[i];
+    ss = wave_sum(ss);
+    if (lane == 0) red[0][wid] = ss;
+    __syncthreads();
+    const float scale =
+        rsqrtf((red[0][0] + red[0][1] + red[0][2] + red[0][3]) / (float)K +
+               eps);
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) {
+        short8 wv = *reinterpret_cast<const short8*>(nw + k0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          // re-quantize so fused == rmsnorm-kernel-then-gemv bit-exactly
+          xr[i * 8 + j] = b2f(f2b(xr[i * 8 + j] * scale * b2f((u16)wv[j])));
+      }
+    }
+  }
+
+  // phase 2: FMA with the prefetched weights (KB<=2) or stream rows with
+  // all rows' loads unrolled together (KB>2) — memory-level parallelism
+  float acc[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) acc[r] = 0.f;
+  if (KB <= 2) {
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+#pragma unroll
+      for (int i = 0; i < KB; ++i) {
+        const int k0 = i * 2048 + t * 8;
+        if (row0 + r < N && k0 < K) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            acc[r] = fmaf(b2f((u16)wpre[r][i][j]), xr[i * 8 + j], acc[r]);
+        }
+      }
+  } else {
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) {
+        short8 wv[ROWS];
+#pragma unroll
+        for (int r = 0; r < ROWS; ++r)
+          if (row0 + r < N) wv[r] = ntload8(W + (size_t)(row0 + r) * K + k0);
+#pragma unroll
+        for (int r = 0; r < ROWS; ++r)
+          if (row0 + r < N) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              acc[r] = fmaf(b2f((u16)wv[r][j]), xr[i * 8 + j], acc[r]);
+          }
+      }
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) {
+    float v = wave_sum(acc[r]);
+    if (lane == 0) red[r][wid] = v;
+  }
+  __syncthreads();
+  if (t < ROWS) {
+    const int row = row0 + t;
+    if (row < N) {
+      float v = red[t][0] + red[t][1] + red[t][2] + red[t][3];
+      if (EPI == 2) {
+        reinterpret_cast<float*>(out)[row] = v;
+      } else if (EPI == 1) {
+        reinterpret_cast<u16*>(out)[row] = f2b(v + b2f(res[row]));
+      } else {
+        reinterpret_cast<u16*>(out)[row] = f2b(v);
+      }
+    }
+  }
+}
+
+// streaming fallback for K > 16384 (e.g. 70B down-proj K=28672): x re-read
+// from L1/L2 per row
+template <int ROWS, int EPI>
+__global__ __launch_bounds__(256) void k_gemv_stream(
+    const u16* __restrict__ W, const u16* __restrict__ x,
+    void* __restrict__ out, const u16* __restrict__ res, int N, int K) {
+  const int t = threadIdx.x;
+  const int row0 = blockIdx.x * ROWS;
+  const int wid = t / WAVE, lane = t % WAVE;
+  float acc[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) acc[r] = 0.f;
+  const int kiter = (K + 2047) / 2048;
+#pragma unroll 1
+  for (int r = 0; r < ROWS; ++r) {
+    const int row = row0 + r;
+    if (row >= N) break;
+    const u16* wr = W + (size_t)row * K;
+    float a = 0.f;
+    for (int i = 0; i < kiter; ++i) {
+      int k0 = i * 2048 + t * 8;
+      if (k0 + 8 <= K) {
+        short8 wv = ntload8(wr + k0);
+        short8 xv = *reinterpret_cast<const short8*>(x + k0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          a = fmaf(b2f((u16)wv[j]), b2f((u16)xv[j]), a);
+      } else {
+        for (int k = k0; k < K; ++k) a = fmaf(b2f(wr[k]), b2f(x[k]), a);
+      }
+    }
+    acc[r] = a;
+  }
+  __shared__ float red[ROWS > 4 ? ROWS : 4][4];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) {
+    float v = wave_sum(acc[r]);
+    if (lane == 0) red[r][wid] = v;
+  }
+  __syncthreads();
+  if (t < ROWS) {
+    const int row = row0 + t;
+    if (row < N) {
+      float v = red[t][0] + red[t][1] + red[t][2] + red[t][3];
+      if (EPI == 2) {
+        reinterpret_cast<float*>(out)[row] = v;
+      } else if (EPI == 1) {
+        reinterpret_cast<u16*>(out)[row] = f2b(v + b2f(res[row]));
+      } else {
+        reinterpret_cast<u16*>(out)[row] = f2b(v);
+      }
+    }
+  }
+}
+
+// fused [rms_norm ->] gate_up GEMV -> silu_mul: block computes
+// out[i] = silu(g_i) * u_i for 8 channels, g_i = dot(W[i,:], xn),
+// u_i = dot(W[i+I,:], xn)  (mlp.rs:21-31 + fused gate_up of mlp.rs:38-46)
+template <int ROWS, bool NORM, int KB>
+__global__ __launch_bounds__(256) void k_gemv_gateup(
+    const u16* __restrict__ W, const u16* __restrict__ x,
+    u16* __restrict__ out, const u16* __restrict__ nw, float eps, int I,
+    int K) {
+  const int t = threadIdx.x;
+  const int c0 = blockIdx.x * ROWS;
+  const int wid = t / WAVE, lane = t % WAVE;
+  __shared__ float redg[8][4], redu[8][4];
+
+  float xr[KB * 8];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    if (k0 < K) {
+      short8 xv = *reinterpret_cast<const short8*>(x + k0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xv[j]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = 0.f;
+    }
+  }
+  if (NORM) {
+    float ss = 0.f;
+#pragma unroll
+    for (int i = 0; i < KB * 8; ++i) ss += xr[i] * xr[i];
+    ss = wave_sum(ss);
+    if (lane == 0) redg[0][wid] = ss;
+    __syncthreads();
+    const float scale =
+        rsqrtf((redg[0][0] + redg[0][1] + redg[0][2] + redg[0][3]) /
+                   (float)K + eps);
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) {
+        short8 wv = *reinterpret_cast<const short8*>(nw + k0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          xr[i * 8 + j] = b2f(f2b(xr[i * 8 + j] * scale * b2f((u16)wv[j])));
+      }
+    }
+  }
+
+  float accg[ROWS], accu[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) accg[r] = accu[r] = 0.f;
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    if (k0 < K) {
+      short8 gv[ROWS], uv[ROWS];
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r)
+        if (c0 + r < I) {
+          gv[r] = ntload8(W + (size_t)(c0 + r) * K + k0);
+          uv[r] = ntload8(W + (size_t)(c0 + r + I) * K + k0);
+        }
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r)
+        if (c0 + r < I) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            accg[r] = fmaf(b2f((u16)gv[r][j]), xr[i * 8 + j], accg[r]);
+            accu[r] = fmaf(b2f((u16)uv[r][j]), xr[i * 8 + j], accu[r]);
+          }
+        }
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) {
+    float g = wave_sum(accg[r]);
+    float u = wave_sum(accu[r]);
+    if (lane == 0) { redg[r][wid] = g; redu[r][wid] = u; }
+  }
+  __syncthreads();
+  if (t < ROWS && c0 + t < I) {
+    float g = redg[t][0] + redg[t][1] + redg[t][2] + redg[t][3];
+    float u = redu[t][0] + redu[t][1] + redu[t][2] + redu[t][3];
+    out[c0 + t] = f2b(g / (1.f + __expf(-g)) * u);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// FP8 GEMV family — same structure as k_gemv_reg but weights are e4m3fn
+// bytes with blockwise 128x128 scale_inv (utils/fp8.rs:42-64), dequantized
+// in-register: HALF the HBM bytes per decode step.  16 weights per 16-B
+// lane load.  KB here = ceil(K/4096) (thread covers 16 elems per iter).
+// ---------------------------------------------------------------------------
+template <int ROWS, int EPI, bool NORM, int KB>
+__global__ __launch_bounds__(256) void k_gemv_fp8(
+    const unsigned char* __restrict__ W, const float* __restrict__ sc,
+    const u16* __restrict__ x, void* __restrict__ out,
+    const u16* __restrict__ res, const u16* __restrict__ nw, float eps,
+    int N, int K, int nkb) {
+  const int t = threadIdx.x;
+  const int row0 = blockIdx.x * ROWS;
+  const int wid = t / WAVE, lane = t % WAVE;
+  __shared__ float red[ROWS > 4 ? ROWS : 4][4];
+
+  // prefetch all weight tiles (stay in flight across the x/norm phase)
+  uint4v wpre[ROWS][KB];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r)
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 4096 + t * 16;
+      if (row0 + r < N && k0 < K)
+        wpre[r][i] = ntload16b(W + (size_t)(row0 + r) * K + k0);
+    }
+
+  // x -> registers (f32), optionally fused rms_norm
+  float xr[KB * 16];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 4096 + t * 16;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      if (k0 < K) {
+        short8 xv = *reinterpret_cast<const short8*>(x + k0 + half * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          xr[i * 16 + half * 8 + j] = b2f((u16)xv[j]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) xr[i * 16 + half * 8 + j] = 0.f;
+      }
+    }
+  }
+  if (NORM) {
+    float ss = 0.f;
+#pragma unroll
+    for (int i = 0; i < KB * 16; ++i) ss += xr[i] * xr[i];
+    ss = wave_sum(ss);
+    if (lane == 0) red[0][wid] = ss;
+    __syncthreads();
+    const float scale =
+        rsqrtf((red[0][0] + red[0][1] + red[0][2] + red[0][3]) / (float)K +
+               eps);
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 4096 + t * 16;
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        if (k0 < K) {
+          short8 wv =
+              *reinterpret_cast<const short8*>(nw + k0 + half * 8);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int ix = i * 16 + half * 8 + j;
+            xr[ix] = b2f(f2b(xr[ix] * scale * b2f((u16)wv[j])));
+          }
+        }
+      }
+    }
+  }
+
+  float acc[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) acc[r] = 0.f;
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r)
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 4096 + t * 16;
+      if (row0 + r < N && k0 < K) {
+        // one scale_inv block covers this thread's 16 k's (k0 % 16 == 0)
+        const float s =
+            sc[(size_t)((row0 + r) / 128) * nkb + (k0 / 128)];
+        float wd[16];
+        f8x16_decode(wpre[r][i], wd);
+        float a = 0.f;
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          a = fmaf(wd[j], xr[i * 16 + j], a);
+        acc[r] = fmaf(a, s, acc[r]);
+      }
+    }
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) {
+    float v = wave_sum(acc[r]);
+    if (lane == 0) red[r][wid] = v;
+  }
+  __syncthreads();
+  if (t < ROWS) {
+    const int row = row0 + t;
+    if (row < N) {
+      float v = red[t][0] + red[t][1] + red[t][2] + red[t][3];
+      if (EPI == 2) {
+        reinterpret_cast<float*>(out)[row] = v;
+      } else if (EPI == 1) {
+        reinterpret_cast<u16*>(out)[row] = f2b(v + b2f(res[row]));
+      } else {
+        reinterpret_cast<u16*>(out)[row] = f2b(v);
+      }
+    }
+  }
+}
+
+// fp8 streaming fallback for K > 16384 (e.g. Qwen3-32B down K=25600)
+template <int ROWS, int EPI>
+__global__ __launch_bounds__(256) void k_gemv_fp8_stream(
+    const unsigned char* __restrict__ W, const float* __restrict__ sc,
+    const u16* __restrict__ x, void* __restrict__ out,
+    const u16* __restrict__ res, int N, int K, int nkb) {
+  const int t = threadIdx.x;
+  const int row0 = blockIdx.x * ROWS;
+  const int wid = t / WAVE, lane = t % WAVE;
+  float acc[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) acc[r] = 0.f;
+  const int kiter = (K + 4095) / 4096;
+#pragma unroll 1
+  for (int i = 0; i < kiter; ++i) {
+    const int k0 = i * 4096 + t * 16;
+    if (k0 >= K) continue;
+    float xv[16];
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      short8 xs = *reinterpret_cast<const short8*>(x + k0 + half * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xv[half * 8 + j] = b2f((u16)xs[j]);
+    }
+    uint4v wv[ROWS];
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+      if (row0 + r < N) wv[r] = ntload16b(W + (size_t)(row0 + r) * K + k0);
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+      if (row0 + r < N) {
+        const float s = sc[(size_t)((row0 + r) / 128) * nkb + (k0 / 128)];
+        float wd[16];
+        f8x16_decode(wv[r], wd);
+        float a = 0.f;
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          a = fmaf(wd[j], xv[j], a);
+        acc[r] = fmaf(a, s, acc[r]);
+      }
+  }
+  __shared__ float red[ROWS > 4 ? ROWS : 4][4];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) {
+    float v = wave_sum(acc[r]);
+    if (lane == 0) red[r][wid] = v;
+  }
+  __syncthreads();
+  if (t < ROWS) {
+    const int row = row0 + t;
+    if (row < N) {
+      float v = red[t][0] + red[t][1] + red[t][2] + red[t][3];
+      if (EPI == 2) {
+        reinterpret_cast<float*>(out)[row] = v;
+      } else if (EPI == 1) {
+        reinterpret_cast<u16*>(out)[row] = f2b(v + b2f(res[row]));
+      } else {
+        reinterpret_cast<u16*>(out)[row] = f2b(v);
+      }
+    }
+  }
+}
+
+// fp8 gate_up + silu_mul (the fused MLP front half, fp8 weights)
+template <int ROWS, bool NORM, int KB>
+__global__ __launch_bounds__(256) void k_gemv_gateup_fp8(
+    const unsigned char* __restrict__ W, const float* __restrict__ sc,
+    const u16* __restrict__ x, u16* __restrict__ out,
+    const u16* __restrict__ nw, float eps, int I, int K, int nkb) {
+  const int t = threadIdx.x;
+  const int c0 = blockIdx.x * ROWS;
+  const int wid = t / WAVE, lane = t % WAVE;
+  __shared__ float redg[8][4], redu[8][4];
+
+  float xr[KB * 16];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 4096 + t * 16;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      if (k0 < K) {
+        short8 xv = *reinterpret_cast<const short8*>(x + k0 + half * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          xr[i * 16 + half * 8 + j] = b2f((u16)xv[j]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) xr[i * 16 + half * 8 + j] = 0.f;
+      }
+    }
+  }
+  if (NORM) {
+    float ss = 0.f;
+#pragma unroll
+    for (int i = 0; i < KB * 16; ++i) ss += xr[i] * xr[i];
+    ss = wave_sum(ss);
+    if (lane == 0) redg[0][wid] = ss;
+    __syncthreads();
+    const float scale =
+        rsqrtf((redg[0][0] + redg[0][1] + redg[0][2] + redg[0][3]) /
+                   (float)K + eps);
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 4096 + t * 16;
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        if (k0 < K) {
+          short8 wv =
+              *reinterpret_cast<const short8*>(nw + k0 + half * 8);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int ix = i * 16 + half * 8 + j;
+            xr[ix] = b2f(f2b(xr[ix] * scale * b2f((u16)wv[j])));
+          }
+        }
+      }
+    }
+  }
+
+  float accg[ROWS], accu[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) accg[r] = accu[r] = 0.f;
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 4096 + t * 16;
+    if (k0 >= K) continue;
+    uint4v gv[ROWS], uv[ROWS];
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+      if (c0 + r < I) {
+        gv[r] = ntload16b(W + (size_t)(c0 + r) * K + k0);
+        uv[r] = ntload16b(W + (size_t)(c0 + r + I) * K + k0);
+      }
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+      if (c0 + r < I) {
+        const float sg = sc[(size_t)((c0 + r) / 128) * nkb + (k0 / 128)];
+        const float su =
+            sc[(size_t)((c0 + r + I) / 128) * nkb + (k0 / 128)];
+        float gd[16], ud[16];
+        f8x16_decode(gv[r], gd);
+        f8x16_decode(uv[r], ud);
+        float g = 0.f, u = 0.f;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          g = fmaf(gd[j], xr[i * 16 + j], g);
+          u = fmaf(ud[j], xr[i * 16 + j], u);
+        }
+        accg[r] = fmaf(g, sg, accg[r]);
+        accu[r] = fmaf(u, su, accu[r]);
+      }
+  }
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) {
+    float g = wave_sum(accg[r]);
+    float u = wave_sum(accu[r]);
+    if (lane == 0) { redg[r][wid] = g; redu[r][wid] = u; }
+  }
+  __syncthreads();
+  if (t < ROWS && c0 + t < I) {
+    float g = redg[t][0] + redg[t][1] + redg[t][2] + redg[t][3];
+    float u = redu[t][0] + redu[t][1] + redu[t][2] + redu[t][3];
+    out[c0 + t] = f2b(g / (1.f + __expf(-g)) * u);
+  }
+}
+
+// fp8 -> bf16 blockwise dequant (prefill path: dequant the layer's weight
+// into a scratch buffer, then run the bf16 MFMA GEMM — cake itself
+// dequantizes at load time, fp8.rs:42-64; we dequant per layer per prefill
+// to keep HBM fp8-resident for decode)
+__global__ void k_dequant_fp8(const unsigned char* __restrict__ W,
+                              const float* __restrict__ sc,
+                              u16* __restrict__ out, int N, int K, int nkb) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  size_t total = (size_t)N * K;
+  for (; i < total; i += stride) {
+    int r = (int)(i / K), k = (int)(i % K);
+    f32x2 v = __builtin_amdgcn_cvt_pk_f32_fp8((u32)W[i], false);
+    out[i] = f2b(v[0] * sc[(size_t)(r / 128) * nkb + k / 128]);
+  }
+}
+
+template <int ROWS, int EPI>
+static void gemv_dispatch_kb(const u16* W, const u16* x, void* out,
+                             const u16* res, const u16* nw, float eps, int N,
+                             int K, hipStream_t s) {
+  dim3 grid((N + ROWS - 1) / ROWS);
+  if (K > 16384) {
+    hipLaunchKernelGGL((k_gemv_stream<ROWS, EPI>), grid, dim3(256), 0, s, W,
+                       x, out, res, N, K);
+    return;
+  }
+#define GEMV_KB(KB)                                                        \
+  do {                                                                     \
+    if (nw)                                                                \
+      hipLaunchKernelGGL((k_gemv_reg<ROWS, EPI, true, KB>), grid,          \
+                         dim3(256), 0, s, W, x, out, res, nw, eps, N, K);  \
+    else                                                                   \
+      hipLaunchKernelGGL((k_gemv_reg<ROWS, EPI, false, KB>), grid,         \
+                         dim3(256), 0, s, W, x, out, res, nw, eps, N, K);  \
+  } while (0)
+  if (K <= 4096) GEMV_KB(2);
+  else if (K <= 8192) GEMV_KB(4);
+  else GEMV_KB(8);
+#undef GEMV_KB
+}
+
+void launch_gemv(const u16* W, const u16* x, void* out, const u16* res,
+                 const u16* nw, float eps, int N, int K, int epi,
+                 hipStream_t s) {
+  // small row-count per block keeps the grid >= ~2048 workgroups on the
+  // usual decode shapes (256 CUs need many blocks to reach HBM peak)
+  static const int env_small = [] {
+    const char* v = getenv("CAKE_GEMV_ROWS_SMALL");
+    return v ? atoi(v) : 0;
+  }();
+  int rows;
+  if (N >= 16384) rows = 8;                 // lm_head: maximal block count
+  else if (env_small) rows = env_small;     // A/B override
+  else if (K >= 8192) rows = 4;             // long rows amortize x reload
+  else rows = (N >= 6144) ? 4 : 2;
+#define GEMV_R(R, EPI) gemv_dispatch_kb<R, EPI>(W, x, out, res, nw, eps, N, K, s)
+#define GEMV_EPI(EPI)                          \
+  do {                                         \
+    if (rows >= 8) GEMV_R(8, EPI);             \
+    else if (rows == 4) GEMV_R(4, EPI);        \
+    else if (rows == 2) GEMV_R(2, EPI);        \
+    else GEMV_R(1, EPI);                       \
+  } while (0)
+  if (epi == 0) GEMV_EPI(0);
+  else if (epi == 1) GEMV_EPI(1);
+  else GEMV_EPI(2);
+#undef GEMV_EPI
+#undef GEMV_R
+}
+void launch_gemv_gateup(const u16* W, const u16* x, u16* out, const u16* nw,
+                        float eps, int I, int K, int rows, hipStream_t s) {
+  if (K > 16384) return;  // guarded at engine create (K = hidden <= 16384)
+#define GU_KB2(ROWS, KB)                                                    \
+  do {                                                                      \
+    dim3 grid((I + ROWS - 1) / ROWS);                                       \
+    if (nw)                                                                 \
+      hipLaunchKernelGGL((k_gemv_gateup<ROWS, true, KB>), grid, dim3(256),  \
+                         0, s, W, x, out, nw, eps, I, K);                   \
+    else                                                                    \
+      hipLaunchKernelGGL((k_gemv_gateup<ROWS, false, KB>), grid, dim3(256), \
+                         0, s, W, x, out, nw, eps, I, K);                   \
+  } while (0)
+#define GU_KB(KB)                                                           \
+  do {                                                                      \
+    if (rows >= 8) GU_KB2(8, KB);                                           \
+    else GU_KB2(4, KB);                                                     \
+  } while (0)
+  if (K <= 4096) GU_KB(2);
+  else if (K <= 8192) GU_KB(4);
+  else GU_KB(8);
+#undef GU_KB
+#undef GU_KB2
+}
+void launch_gemv_fp8(const unsigned char* W, const float* sc, const u16* x,
+                     void* out, const u16* res, const u16* nw, float eps,
+                     int N, int K, int epi, hipStream_t s) {
+  const int nkb = (K + 127) / 128;
+  static const int env_rows = [] {
+    const char* v = getenv("CAKE_FP8_ROWS");
+    return v ? atoi(v) : 0;
+  }();
+  const int rows = N >= 16384 ? 8 : (env_rows ? env_rows : 4);
+#define F8_KB(R, EPI, KB)                                                   \
+  do {                                                                      \
+    dim3 grid((N + R - 1) / R);                                             \
+    if (nw)                                                                 \
+      hipLaunchKernelGGL((k_gemv_fp8<R, EPI, true, KB>), grid, dim3(256),   \
+                         0, s, W, sc, x, out, res, nw, eps, N, K, nkb);     \
+    else                                                                    \
+      hipLaunchKernelGGL((k_gemv_fp8<R, EPI, false, KB>), grid, dim3(256),  \
+                         0, s, W, sc, x, out, res, nw, eps, N, K, nkb);     \
+  } while (0)
+#define F8_R(R, EPI)                                                        \
+  do {                                                                      \
+    if (K > 16384) {                                                        \
+      dim3 grid((N + R - 1) / R);                                           \
+      hipLaunchKernelGGL((k_gemv_fp8_stream<R, EPI>), grid, dim3(256), 0,   \
+                         s, W, sc, x, out, res, N, K, nkb);                 \
+    } else if (K <= 4096) F8_KB(R, EPI, 1);                                 \
+    else if (K <= 8192) F8_KB(R, EPI, 2);                                   \
+    else F8_KB(R, EPI, 4);                                                  \
+  } while (0)
+  if (epi == 0) { if (rows == 8) F8_R(8, 0); else F8_R(4, 0); }
+  else if (epi == 1) { if (rows == 8) F8_R(8, 1); else F8_R(4, 1); }
+  else { if (rows == 8) F8_R(8, 2); else F8_R(4, 2); }
+#undef F8_R
+#undef F8_KB
+}
+void launch_gemv_gateup_fp8(const unsigned char* W, const float* sc,
+                            const u16* x, u16* out, const u16* nw, float eps,
+                            int I, int K, hipStream_t s) {
+  const int nkb = (K + 127) / 128;
+  dim3 grid((I + 3) / 4);
+#define GU8_KB(KB)                                                          \
+  do {                                                                      \
+    if (nw)                                                                 \
+      hipLaunchKernelGGL((k_gemv_gateup_fp8<4, true, KB>), grid, dim3(256), \
+                         0, s, W, sc, x, out, nw, eps, I, K, nkb);          \
+    else                                                                    \
+      hipLaunchKernelGGL((k_gemv_gateup_fp8<4, false, KB>), grid,           \
+                         dim3(256), 0, s, W, sc, x, out, nw, eps, I, K,     \
+                         nkb);                                              \
+  } while (0)
+  if (K <= 4096) GU8_KB(1);
+  else if (K <= 8192) GU8_KB(2);
+  else GU8_KB(4);
+#undef GU8_KB
+}
+void launch_dequant_fp8(const unsigned char* W, const float* sc, u16* out,
+                        int N, int K, hipStream_t s) {
+  const int nkb = (K + 127) / 128;
+  hipLaunchKernelGGL(k_dequant_fp8, dim3(4096), dim3(256), 0, s, W, sc, out,
+                     N, K, nkb);
+}

@@ -1,0 +1,337 @@
+#include "kernels_common.h"
+#include "kernels.h"
+
+
+// ---------------------------------------------------------------------------
+// conversion / fill
+// ---------------------------------------------------------------------------
+__global__ void k_f32_to_bf16(const float* __restrict__ in,
+                              u16* __restrict__ out, size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = f2b(in[i]);
+}
+__global__ void k_bf16_to_f32(const u16* __restrict__ in,
+                              float* __restrict__ out, size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = b2f(in[i]);
+}
+
+// splitmix64-based uniform(-1,1)*scale fill (deterministic per element)
+__global__ void k_fill_random(u16* __restrict__ out, size_t n, uint64_t seed,
+                              float scale) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    uint64_t z = seed + 0x9e3779b97f4a7c15ull * (i + 1);
+    z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
+    z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
+    z ^= z >> 31;
+    float u = (float)(z >> 40) * (1.0f / 8388608.0f) - 1.0f;  // [-1, 1)
+    out[i] = f2b(u * scale);
+  }
+}
+// random e4m3fn bytes for synthetic fp8 weights: NaN encodings (0x7f/0xff)
+// are remapped into the largest normal
+__global__ void k_fill_random_u8(unsigned char* __restrict__ out, size_t n,
+                                 uint64_t seed) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    uint64_t z = seed + 0x9e3779b97f4a7c15ull * (i + 1);
+    z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
+    z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
+    z ^= z >> 31;
+    unsigned char b = (unsigned char)(z >> 56);
+    if ((b & 0x7f) == 0x7f) b ^= 0x08;  // avoid NaN encodings
+    out[i] = b;
+  }
+}
+__global__ void k_fill_const(u16* __restrict__ out, size_t n, float v) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  u16 b = f2b(v);
+  for (; i < n; i += stride) out[i] = b;
+}
+
+// ---------------------------------------------------------------------------
+// rms_norm (backends/mod.rs:244-246): out = x * w / sqrt(mean(x^2) + eps)
+// one block per row; bf16x8 vector loads (guide G13); f32 accumulate
+// ---------------------------------------------------------------------------
+// rows are addressed as (outer, inner): row r -> offset
+// (r/inner)*outer_stride + (r%inner)*cols.  Plain contiguous rows use
+// inner = nrows, outer_stride = 0.  The (outer, inner) form handles the
+// per-head QK-norm rows inside the strided qkv buffer
+// (attention.rs:202-215: norm over head_dim after reshape).
+__global__ void k_rmsnorm(const u16* __restrict__ x, const u16* __restrict__ w,
+                          u16* __restrict__ out, int cols, float eps,
+                          int inner, size_t outer_stride) {
+  const int row = blockIdx.x;
+  const size_t off =
+      (size_t)(row / inner) * outer_stride + (size_t)(row % inner) * cols;
+  const u16* xr = x + off;
+  u16* outr = out + off;
+  float ss = 0.f;
+  const int t = threadIdx.x;
+  const int nvec = cols / 8;
+  for (int i = t; i < nvec; i += blockDim.x) {
+    short8 v = *reinterpret_cast<const short8*>(xr + i * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = b2f((u16)v[j]);
+      ss += f * f;
+    }
+  }
+  for (int i = nvec * 8 + t; i < cols; i += blockDim.x) {
+    float f = b2f(xr[i]);
+    ss += f * f;
+  }
+  ss = wave_sum(ss);
+  __shared__ float red[16];
+  const int wid = t / WAVE, lane = t % WAVE;
+  if (lane == 0) red[wid] = ss;
+  __syncthreads();
+  const int nw = blockDim.x / WAVE;
+  float tot = 0.f;
+#pragma unroll
+  for (int i = 0; i < 16; ++i)
+    if (i < nw) tot += red[i];
+  const float scale = rsqrtf(tot / (float)cols + eps);
+  for (int i = t; i < nvec; i += blockDim.x) {
+    short8 v = *reinterpret_cast<const short8*>(xr + i * 8);
+    short8 wv = *reinterpret_cast<const short8*>(w + i * 8);
+    short8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = (short)f2b(b2f((u16)v[j]) * scale * b2f((u16)wv[j]));
+    *reinterpret_cast<short8*>(outr + i * 8) = o;
+  }
+  for (int i = nvec * 8 + t; i < cols; i += blockDim.x)
+    outr[i] = f2b(b2f(xr[i]) * scale * b2f(w[i]));
+}
+
+// ---------------------------------------------------------------------------
+// silu_mul (ops.cu:101-138): out = gate * sigmoid(gate) * up
+// ---------------------------------------------------------------------------
+__global__ void k_silu_mul(const u16* __restrict__ gate,
+                           const u16* __restrict__ up, u16* __restrict__ out,
+                           size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float g = b2f(gate[i]);
+    float u = b2f(up[i]);
+    out[i] = f2b(g / (1.f + __expf(-g)) * u);
+  }
+}
+// strided variant for the fused gate_up buffer (S, 2I): gate = row[0:I],
+// up = row[I:2I]  (mlp.rs:21-31 narrow semantics)
+__global__ void k_silu_mul_rows(const u16* __restrict__ gu,
+                                u16* __restrict__ out, int S, int I) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  size_t n = (size_t)S * I;
+  for (; i < n; i += stride) {
+    int s = (int)(i / I), c = (int)(i % I);
+    float g = b2f(gu[(size_t)s * 2 * I + c]);
+    float u = b2f(gu[(size_t)s * 2 * I + I + c]);
+    out[i] = f2b(g / (1.f + __expf(-g)) * u);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// embedding gather (backends/mod.rs:513-528)
+// ---------------------------------------------------------------------------
+__global__ void k_embed_token(const u16* __restrict__ embed,
+                              const u32* __restrict__ tok,
+                              u16* __restrict__ x, int H) {
+  const u16* src = embed + (size_t)(*tok) * H;
+  const int nv = H / 8;
+  for (int i = threadIdx.x; i < nv; i += blockDim.x)
+    *reinterpret_cast<short8*>(x + i * 8) =
+        *reinterpret_cast<const short8*>(src + i * 8);
+  for (int i = nv * 8 + threadIdx.x; i < H; i += blockDim.x) x[i] = src[i];
+}
+__global__ void k_embed_rows(const u16* __restrict__ embed,
+                             const u32* __restrict__ ids,
+                             u16* __restrict__ x, int H) {
+  const int s = blockIdx.x;
+  const u16* src = embed + (size_t)ids[s] * H;
+  u16* dst = x + (size_t)s * H;
+  for (int i = threadIdx.x; i < H; i += blockDim.x) dst[i] = src[i];
+}
+
+// standalone rope for the op-level parity surface: x (BH, S, D) bf16 in
+// place, cos/sin (S, D/2) f32 (backends/mod.rs:444-482 layout)
+__global__ void k_rope_simple(u16* __restrict__ x,
+                              const float* __restrict__ cost,
+                              const float* __restrict__ sint, int s, int d) {
+  const int row = blockIdx.x;  // (bh, s) flattened
+  const int si = row % s;
+  const int half = d / 2;
+  u16* xr = x + (size_t)row * d;
+  const float* c = cost + (size_t)si * half;
+  const float* sn = sint + (size_t)si * half;
+  for (int i = threadIdx.x; i < half; i += blockDim.x) {
+    float x1 = b2f(xr[i]), x2 = b2f(xr[i + half]);
+    xr[i] = f2b(x1 * c[i] - x2 * sn[i]);
+    xr[i + half] = f2b(x2 * c[i] + x1 * sn[i]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// argmax (greedy sampling, text_model.rs:104): first-index tie-break to
+// match np.argmax / candle ArgMax.  Two passes over f32 logits.
+// ---------------------------------------------------------------------------
+__global__ void k_argmax_part(const float* __restrict__ logits, int n,
+                              float* __restrict__ pval, int* __restrict__ pidx,
+                              int nparts) {
+  const int part = blockIdx.x;
+  const int span = (n + nparts - 1) / nparts;
+  const int start = part * span, end = min(start + span, n);
+  float best = -INFINITY;
+  int bidx = 0x7fffffff;
+  for (int i = start + (int)threadIdx.x; i < end; i += blockDim.x) {
+    float v = logits[i];
+    if (v > best || (v == best && i < bidx)) { best = v; bidx = i; }
+  }
+  // wave+block reduce keeping first index on ties
+  __shared__ float sv[256];
+  __shared__ int si[256];
+  sv[threadIdx.x] = best;
+  si[threadIdx.x] = bidx;
+  __syncthreads();
+  for (int off = blockDim.x / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      float ov = sv[threadIdx.x + off];
+      int oi = si[threadIdx.x + off];
+      if (ov > sv[threadIdx.x] ||
+          (ov == sv[threadIdx.x] && oi < si[threadIdx.x])) {
+        sv[threadIdx.x] = ov;
+        si[threadIdx.x] = oi;
+      }
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) { pval[part] = sv[0]; pidx[part] = si[0]; }
+}
+
+// Final reduce; also appends the winning token to the output ring and
+// advances the device position (end-of-decode-step bookkeeping).
+__global__ void k_argmax_fin(const float* __restrict__ pval,
+                             const int* __restrict__ pidx, int nparts,
+                             u32* __restrict__ tok, int* __restrict__ pos,
+                             u32* __restrict__ ring, int* __restrict__ step,
+                             int advance_pos) {
+  __shared__ float sv[256];
+  __shared__ int si[256];
+  float best = -INFINITY;
+  int bidx = 0x7fffffff;
+  for (int i = threadIdx.x; i < nparts; i += blockDim.x) {
+    float v = pval[i];
+    if (v > best || (v == best && pidx[i] < bidx)) { best = v; bidx = pidx[i]; }
+  }
+  sv[threadIdx.x] = best;
+  si[threadIdx.x] = bidx;
+  __syncthreads();
+  for (int off = blockDim.x / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      float ov = sv[threadIdx.x + off];
+      int oi = si[threadIdx.x + off];
+      if (ov > sv[threadIdx.x] ||
+          (ov == sv[threadIdx.x] && oi < si[threadIdx.x])) {
+        sv[threadIdx.x] = ov;
+        si[threadIdx.x] = oi;
+      }
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    *tok = (u32)si[0];
+    if (ring) {
+      ring[*step] = (u32)si[0];
+      *step += 1;
+    }
+    if (advance_pos) *pos += 1;
+  }
+}
+
+__global__ void k_advance_pos(int* __restrict__ pos, int by) {
+  if (threadIdx.x == 0) *pos += by;
+}
+
+void launch_bf16_to_f32(const u16* in, float* out, size_t n, hipStream_t s) {
+  int blocks = (int)min((n + 255) / 256, (size_t)2048);
+  hipLaunchKernelGGL(k_bf16_to_f32, dim3(blocks), dim3(256), 0, s, in, out, n);
+}
+void launch_fill_random(u16* out, size_t n, uint64_t seed, float scale,
+                        hipStream_t s) {
+  hipLaunchKernelGGL(k_fill_random, dim3(2048), dim3(256), 0, s, out, n, seed,
+                     scale);
+}
+void launch_fill_random_u8(unsigned char* out, size_t n, uint64_t seed,
+                           hipStream_t s) {
+  hipLaunchKernelGGL(k_fill_random_u8, dim3(2048), dim3(256), 0, s, out, n,
+                     seed);
+}
+void launch_fill_const(u16* out, size_t n, float v, hipStream_t s) {
+  hipLaunchKernelGGL(k_fill_const, dim3(512), dim3(256), 0, s, out, n, v);
+}
+void launch_rmsnorm(const u16* x, const u16* w, u16* out, int rows, int cols,
+                    float eps, hipStream_t s) {
+  hipLaunchKernelGGL(k_rmsnorm, dim3(rows), dim3(256), 0, s, x, w, out, cols,
+                     eps, rows, (size_t)0);
+}
+void launch_rmsnorm_strided(const u16* x, const u16* w, u16* out, int outer,
+                            int inner, size_t outer_stride, int cols,
+                            float eps, hipStream_t s) {
+  hipLaunchKernelGGL(k_rmsnorm, dim3(outer * inner), dim3(256), 0, s, x, w,
+                     out, cols, eps, inner, outer_stride);
+}
+void launch_silu_mul_rows(const u16* gu, u16* out, int S, int I,
+                          hipStream_t s) {
+  size_t n = (size_t)S * I;
+  int blocks = (int)min((n + 255) / 256, (size_t)4096);
+  hipLaunchKernelGGL(k_silu_mul_rows, dim3(blocks), dim3(256), 0, s, gu, out,
+                     S, I);
+}
+void launch_silu_mul(const u16* g, const u16* u, u16* out, size_t n,
+                     hipStream_t s) {
+  int blocks = (int)min((n + 255) / 256, (size_t)4096);
+  hipLaunchKernelGGL(k_silu_mul, dim3(blocks), dim3(256), 0, s, g, u, out, n);
+}
+void launch_embed_token(const u16* embed, const u32* tok, u16* x, int H,
+                        hipStream_t s) {
+  hipLaunchKernelGGL(k_embed_token, dim3(1), dim3(256), 0, s, embed, tok, x, H);
+}
+void launch_embed_rows(const u16* embed, const u32* ids, u16* x, int S, int H,
+                       hipStream_t s) {
+  hipLaunchKernelGGL(k_embed_rows, dim3(S), dim3(256), 0, s, embed, ids, x, H);
+}
+void launch_rope_simple(u16* x, const float* cost, const float* sint, int bh,
+                        int s, int d, hipStream_t st) {
+  hipLaunchKernelGGL(k_rope_simple, dim3(bh * s), dim3(64), 0, st, x, cost,
+                     sint, s, d);
+}
+void launch_argmax(const float* logits, int n, float* pval, int* pidx,
+                   u32* tok, int* pos, u32* ring, int* step, int advance_pos,
+                   hipStream_t s) {
+  const int nparts = 256;
+  hipLaunchKernelGGL(k_argmax_part, dim3(nparts), dim3(256), 0, s, logits, n,
+                     pval, pidx, nparts);
+  hipLaunchKernelGGL(k_argmax_fin, dim3(1), dim3(256), 0, s, pval, pidx,
+                     nparts, tok, pos, ring, step, advance_pos);
+}
+void launch_advance_pos(int* pos, int by, hipStream_t s) {
+  hipLaunchKernelGGL(k_advance_pos, dim3(1), dim3(64), 0, s, pos, by);
+}
+
+// ---------------------------------------------------------------------------
+// launch wrappers (host side) — declared in kernels.h, used by engine.hip
+// ---------------------------------------------------------------------------
+
+void launch_f32_to_bf16(const float* in, u16* out, size_t n, hipStream_t s) {
+  int blocks = (int)min((n + 255) / 256, (size_t)2048);
+  hipLaunchKernelGGL(k_f32_to_bf16, dim3(blocks), dim3(256), 0, s, in, out, n);
+}
