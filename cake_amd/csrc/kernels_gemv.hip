@@ -14,14 +14,11 @@
 // launcher so the grid has >= ~2048 workgroups (256 CUs want many blocks).
 // EPI: 0 = bf16 out; 1 = bf16 out + residual add; 2 = f32 out (logits).
 // ---------------------------------------------------------------------------
-template <int ROWS, int EPI, bool NORM, int KB, bool GUARD>
+template <int ROWS, int EPI, bool NORM, int KB>
 __global__ __launch_bounds__(256) void k_gemv_reg(
     const u16* __restrict__ W, const u16* __restrict__ x,
     void* __restrict__ out, const u16* __restrict__ res,
     const u16* __restrict__ nw, float eps, int N, int K) {
-  // GUARD=false fast path (K % 2048 == 0 and N % ROWS == 0): bounds
-  // collapse to wave-uniform i*2048 < K, so no exec-mask waterfalls around
-  // the loads
   const int t = threadIdx.x;
   const int row0 = blockIdx.x * ROWS;
   const int wid = t / WAVE, lane = t % WAVE;
@@ -37,7 +34,7 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
 #pragma unroll
       for (int i = 0; i < KB; ++i) {
         const int k0 = i * 2048 + t * 8;
-        if (GUARD ? (row0 + r < N && k0 < K) : (i * 2048 < K))
+        if (row0 + r < N && k0 < K)
           wpre[r][i] = ntload8(W + (size_t)(row0 + r) * K + k0);
       }
   }
@@ -47,7 +44,7 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
     const int k0 = i * 2048 + t * 8;
-    if (GUARD ? (k0 < K) : (i * 2048 < K)) {
+    if (k0 < K) {
       short8 xv = *reinterpret_cast<const short8*>(x + k0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xv[j]);
@@ -70,7 +67,7 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
 #pragma unroll
     for (int i = 0; i < KB; ++i) {
       const int k0 = i * 2048 + t * 8;
-      if (GUARD ? (k0 < K) : (i * 2048 < K)) {
+      if (k0 < K) {
         short8 wv = *reinterpret_cast<const short8*>(nw + k0);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
@@ -91,7 +88,7 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
 #pragma unroll
       for (int i = 0; i < KB; ++i) {
         const int k0 = i * 2048 + t * 8;
-        if (GUARD ? (row0 + r < N && k0 < K) : (i * 2048 < K)) {
+        if (row0 + r < N && k0 < K) {
 #pragma unroll
           for (int j = 0; j < 8; ++j)
             acc[r] = fmaf(b2f((u16)wpre[r][i][j]), xr[i * 8 + j], acc[r]);
@@ -101,15 +98,14 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
 #pragma unroll
     for (int i = 0; i < KB; ++i) {
       const int k0 = i * 2048 + t * 8;
-      if (GUARD ? (k0 < K) : (i * 2048 < K)) {
+      if (k0 < K) {
         short8 wv[ROWS];
 #pragma unroll
         for (int r = 0; r < ROWS; ++r)
-          if (!GUARD || row0 + r < N)
-            wv[r] = ntload8(W + (size_t)(row0 + r) * K + k0);
+          if (row0 + r < N) wv[r] = ntload8(W + (size_t)(row0 + r) * K + k0);
 #pragma unroll
         for (int r = 0; r < ROWS; ++r)
-          if (!GUARD || row0 + r < N) {
+          if (row0 + r < N) {
 #pragma unroll
             for (int j = 0; j < 8; ++j)
               acc[r] = fmaf(b2f((u16)wv[r][j]), xr[i * 8 + j], acc[r]);
@@ -592,25 +588,19 @@ static void gemv_dispatch_kb(const u16* W, const u16* x, void* out,
                        x, out, res, N, K);
     return;
   }
-  const bool full = (K % 2048 == 0) && (N % ROWS == 0);
-#define GEMV_KB1(KB, NORM, GUARD)                                          \
-  hipLaunchKernelGGL((k_gemv_reg<ROWS, EPI, NORM, KB, GUARD>), grid,       \
-                     dim3(256), 0, s, W, x, out, res, nw, eps, N, K)
 #define GEMV_KB(KB)                                                        \
   do {                                                                     \
-    if (nw) {                                                              \
-      if (full) GEMV_KB1(KB, true, false);                                 \
-      else GEMV_KB1(KB, true, true);                                       \
-    } else {                                                               \
-      if (full) GEMV_KB1(KB, false, false);                                \
-      else GEMV_KB1(KB, false, true);                                      \
-    }                                                                      \
+    if (nw)                                                                \
+      hipLaunchKernelGGL((k_gemv_reg<ROWS, EPI, true, KB>), grid,          \
+                         dim3(256), 0, s, W, x, out, res, nw, eps, N, K);  \
+    else                                                                   \
+      hipLaunchKernelGGL((k_gemv_reg<ROWS, EPI, false, KB>), grid,         \
+                         dim3(256), 0, s, W, x, out, res, nw, eps, N, K);  \
   } while (0)
   if (K <= 4096) GEMV_KB(2);
   else if (K <= 8192) GEMV_KB(4);
   else GEMV_KB(8);
 #undef GEMV_KB
-#undef GEMV_KB1
 }
 
 void launch_gemv(const u16* W, const u16* x, void* out, const u16* res,
