@@ -62,6 +62,8 @@ def _load_lib():
     lib.cake_hip_kernel_stats.argtypes = [p, ctypes.c_char_p, c]
     lib.cake_hip_stats_reset.argtypes = [p]
     lib.cake_hip_set_stats.argtypes = [p, c]
+    lib.cake_hip_set_sampling.argtypes = [p, ctypes.c_float,
+                                          ctypes.c_uint64]
     lib.cake_hip_sync.argtypes = [p]
     return lib
 
@@ -235,6 +237,12 @@ class Engine:
 
     def sync(self):
         _check(_lib.cake_hip_sync(self._h))
+
+    def set_sampling(self, temperature, seed=299792458):
+        """temperature <= 0 = greedy ArgMax (cake's convention,
+        text_model.rs:104); > 0 = on-GPU Gumbel-argmax sampling."""
+        _check(_lib.cake_hip_set_sampling(
+            self._h, ctypes.c_float(temperature), seed))
 
     def set_stats(self, enabled):
         _check(_lib.cake_hip_set_stats(self._h, 1 if enabled else 0))

@@ -240,6 +240,8 @@ struct cake_engine {
   int host_pos = 0;
 
   int nchunk = 8;   // split-KV chunks for decode attention (CAKE_NCHUNK)
+  float inv_temp = 0.f;        // 0 = greedy ArgMax (text_model.rs:104)
+  uint64_t sample_seed = 299792458ull;  // cake default seed (lib.rs:180)
   int gu_rows = 4;  // gate_up channels per block (CAKE_GU_ROWS)
 
   hipStream_t stream = nullptr;
@@ -474,10 +476,12 @@ static void enqueue_head_sample(cake_engine* e, int S, int advance_by) {
   }
   if (advance_by > 1)
     launch_advance_pos(e->dev_pos, advance_by - 1, e->stream);
-  {  // greedy ArgMax + ring append + pos++ (text_model.rs:104)
+  {  // greedy ArgMax / Gumbel sampling + ring append + pos++
+     // (text_model.rs:102-118)
     StatScope ss(e, "argmax", (double)V * 4, 0);
     launch_argmax(e->logits, V, e->pval, e->pidx, e->dev_tok, e->dev_pos,
-                  e->ring, e->dev_step, 1, e->stream);
+                  e->ring, e->dev_step, 1, e->inv_temp, e->sample_seed,
+                  e->stream);
   }
 }
 
@@ -1331,6 +1335,22 @@ extern "C" int cake_hip_op_rope(int b, int h, int s, int d, const float* x,
 // ---------------------------------------------------------------------------
 // stats / misc
 // ---------------------------------------------------------------------------
+// Sampling config (create_logits_processor, text_model.rs:102-118):
+// temperature <= 0 -> greedy ArgMax; > 0 -> Gumbel-argmax at that
+// temperature.  Changing it invalidates a captured decode graph.
+extern "C" int cake_hip_set_sampling(cake_engine* e, float temperature,
+                                     uint64_t seed) {
+  HIP_TRY(hipSetDevice(e->device));
+  HIP_TRY(hipStreamSynchronize(e->stream));
+  e->inv_temp = temperature > 0.f ? 1.0f / temperature : 0.f;
+  e->sample_seed = seed;
+  if (e->graph) {
+    hipGraphExecDestroy(e->graph);
+    e->graph = nullptr;
+  }
+  return 0;
+}
+
 extern "C" int cake_hip_set_stats(cake_engine* e, int enabled) {
   e->st.on = enabled != 0;
   return 0;

@@ -62,7 +62,7 @@ void launch_rope_simple(u16* x, const float* cost, const float* sint, int bh,
                         int s, int d, hipStream_t st);
 void launch_argmax(const float* logits, int n, float* pval, int* pidx,
                    u32* tok, int* pos, u32* ring, int* step, int advance_pos,
-                   hipStream_t s);
+                   float inv_temp, uint64_t seed, hipStream_t s);
 void launch_advance_pos(int* pos, int by, hipStream_t s);
 void launch_gemm(const u16* A, const u16* W, u16* C, const u16* res, int M,
                  int N, int K, int epi, hipStream_t s);

@@ -184,16 +184,35 @@ __global__ void k_rope_simple(u16* __restrict__ x,
 // argmax (greedy sampling, text_model.rs:104): first-index tie-break to
 // match np.argmax / candle ArgMax.  Two passes over f32 logits.
 // ---------------------------------------------------------------------------
+// Gumbel-argmax sampling (the trick cake uses for temperature > 0,
+// text_model.rs:108-111 / candle Sampling::GumbelSoftmax): sample =
+// argmax(logits / T + G), G = -log(-log(u)).  inv_temp == 0 selects plain
+// greedy ArgMax (temperature <= 0 semantics, text_model.rs:104).  The
+// uniform stream is a splitmix64 hash of (seed, step, index) — deterministic
+// and graph-replayable (step read from the device step counter); NOT
+// bit-matched to candle's StdRng (documented in DESIGN.md).
 __global__ void k_argmax_part(const float* __restrict__ logits, int n,
                               float* __restrict__ pval, int* __restrict__ pidx,
-                              int nparts) {
+                              int nparts, float inv_temp, uint64_t seed,
+                              const int* __restrict__ step) {
   const int part = blockIdx.x;
   const int span = (n + nparts - 1) / nparts;
   const int start = part * span, end = min(start + span, n);
+  const uint64_t base =
+      seed + 0x9e3779b97f4a7c15ull * (uint64_t)(*step + 1);
   float best = -INFINITY;
   int bidx = 0x7fffffff;
   for (int i = start + (int)threadIdx.x; i < end; i += blockDim.x) {
     float v = logits[i];
+    if (inv_temp > 0.f) {
+      uint64_t z = base + 0xbf58476d1ce4e5b9ull * (uint64_t)(i + 1);
+      z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
+      z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
+      z ^= z >> 31;
+      // u in (0, 1): top 24 bits, +1 to avoid exactly 0
+      float u = ((float)(z >> 40) + 1.0f) * (1.0f / 16777217.0f);
+      v = v * inv_temp + (-__logf(-__logf(u)));
+    }
     if (v > best || (v == best && i < bidx)) { best = v; bidx = i; }
   }
   // wave+block reduce keeping first index on ties
@@ -316,10 +335,10 @@ void launch_rope_simple(u16* x, const float* cost, const float* sint, int bh,
 }
 void launch_argmax(const float* logits, int n, float* pval, int* pidx,
                    u32* tok, int* pos, u32* ring, int* step, int advance_pos,
-                   hipStream_t s) {
+                   float inv_temp, uint64_t seed, hipStream_t s) {
   const int nparts = 256;
   hipLaunchKernelGGL(k_argmax_part, dim3(nparts), dim3(256), 0, s, logits, n,
-                     pval, pidx, nparts);
+                     pval, pidx, nparts, inv_temp, seed, step);
   hipLaunchKernelGGL(k_argmax_fin, dim3(1), dim3(256), 0, s, pval, pidx,
                      nparts, tok, pos, ring, step, advance_pos);
 }

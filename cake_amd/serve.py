@@ -7,9 +7,9 @@ Mirrors cake's API surface (cake-core/src/cake/sharding/api/mod.rs:66-110):
   GET  /api/v1/topology       (api/mod.rs:104-107)
 
 Generation is the same hot loop the bench measures (Master::generate_text,
-master.rs:109-171): greedy ArgMax decode (temperature <= 0 semantics,
-text_model.rs:104); non-greedy sampling is out of round-1 scope and
-`temperature` is accepted but treated as 0.
+master.rs:109-171): greedy ArgMax when temperature <= 0, on-GPU
+Gumbel-argmax sampling otherwise (cake's own temperature-sampling trick,
+text_model.rs:102-118); top-k/top-p are round-2 items.
 
 Tokenization: pass a `tokenizers.Tokenizer` (tokenizer.json) for text
 prompts; without one, requests supply `prompt_token_ids` directly (the
@@ -107,6 +107,9 @@ def create_app(engine, model_name="cake-amd", tokenizer=None, eos_ids=(),
     async def run_request(body, kind):
         max_tokens = int(body.get("max_tokens", 128))
         stream = bool(body.get("stream", False))
+        if hasattr(engine, "set_sampling"):
+            engine.set_sampling(float(body.get("temperature", 0.0)),
+                                int(body.get("seed", 299792458)))
         if "prompt_token_ids" in body:
             ids = [int(t) for t in body["prompt_token_ids"]]
         elif kind == "chat":

@@ -145,6 +145,12 @@ int cake_hip_op_rope(int b, int h, int s, int d, const float *x,
  * kernel-family hipEvent timing + algorithmic bytes, collected while
  * CAKE_HIP_STATS was set (SURVEY.md §5 metrics row). */
 int cake_hip_kernel_stats(cake_engine *e, char *buf, int cap);
+
+/* Sampling config (create_logits_processor, text_model.rs:102-118):
+ * temperature <= 0 => greedy ArgMax; > 0 => Gumbel-argmax at that
+ * temperature (the on-GPU sampling trick cake uses, text_model.rs:108-111).
+ * The noise stream is seeded and deterministic per (seed, step). */
+int cake_hip_set_sampling(cake_engine *e, float temperature, uint64_t seed);
 int cake_hip_stats_reset(cake_engine *e);
 int cake_hip_set_stats(cake_engine *e, int enabled);
 
