@@ -517,24 +517,17 @@ __global__ __launch_bounds__(256) void k_gemv_gateup_fp8(
   float accg[ROWS], accu[ROWS];
 #pragma unroll
   for (int r = 0; r < ROWS; ++r) accg[r] = accu[r] = 0.f;
-  // prefetch ALL weight tiles first (both streams, both k-iterations) so
-  // no load waits behind a decode chain
-  uint4v gpre[ROWS][KB], upre[ROWS][KB];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
     const int k0 = i * 4096 + t * 16;
     if (k0 >= K) continue;
+    uint4v gv[ROWS], uv[ROWS];
 #pragma unroll
     for (int r = 0; r < ROWS; ++r)
       if (c0 + r < I) {
-        gpre[r][i] = ntload16b(W + (size_t)(c0 + r) * K + k0);
-        upre[r][i] = ntload16b(W + (size_t)(c0 + r + I) * K + k0);
+        gv[r] = ntload16b(W + (size_t)(c0 + r) * K + k0);
+        uv[r] = ntload16b(W + (size_t)(c0 + r + I) * K + k0);
       }
-  }
-#pragma unroll
-  for (int i = 0; i < KB; ++i) {
-    const int k0 = i * 4096 + t * 16;
-    if (k0 >= K) continue;
 #pragma unroll
     for (int r = 0; r < ROWS; ++r)
       if (c0 + r < I) {
@@ -542,8 +535,8 @@ __global__ __launch_bounds__(256) void k_gemv_gateup_fp8(
         const float su =
             sc[(size_t)((c0 + r + I) / 128) * nkb + (k0 / 128)];
         float gd[16], ud[16];
-        f8x16_decode(gpre[r][i], gd);
-        f8x16_decode(upre[r][i], ud);
+        f8x16_decode(gv[r], gd);
+        f8x16_decode(uv[r], ud);
         float g = 0.f, u = 0.f;
 #pragma unroll
         for (int j = 0; j < 16; ++j) {
