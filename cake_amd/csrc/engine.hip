@@ -28,6 +28,7 @@
 #include <cstdio>
 #include <cstring>
 #include <map>
+#include <set>
 #include <string>
 #include <vector>
 
@@ -252,6 +253,10 @@ struct cake_engine {
   int rank = 0, world = 1;
 
   Stats st;
+
+  // multi-file (sharded) checkpoint loading state
+  bool loading_multi = false;
+  std::set<std::string> missing;
 };
 
 static int dev_alloc(void** p, size_t bytes) {
@@ -751,7 +756,89 @@ static int upload_weight(cake_engine* e, u16* dst, const char* filebase,
   return 0;
 }
 
+// Load one safetensors file's relevant tensors (helper for single- and
+// multi-file checkpoints).
+static int load_safetensors_file(cake_engine* e, const char* path);
+
+// Entry point: `path` may be (a) one .safetensors file, (b) a
+// model.safetensors.index.json (sharded checkpoint — the layout cake's
+// VarBuilder reads, utils/mod.rs:251-370), or (c) a directory containing
+// either.
 extern "C" int cake_hip_load_safetensors(cake_engine* e, const char* path) {
+  std::string p(path);
+  auto ends_with = [&](const std::string& s, const char* suf) {
+    size_t n = strlen(suf);
+    return s.size() >= n && s.compare(s.size() - n, n, suf) == 0;
+  };
+  // directory? try model.safetensors.index.json then model.safetensors
+  if (!ends_with(p, ".safetensors") && !ends_with(p, ".json")) {
+    std::string idx = p + "/model.safetensors.index.json";
+    FILE* t = fopen(idx.c_str(), "rb");
+    if (t) {
+      fclose(t);
+      p = idx;
+    } else {
+      p = p + "/model.safetensors";
+    }
+  }
+  if (ends_with(p, ".json")) {
+    // sharded checkpoint: weight_map = {tensor: file}; load each distinct
+    // file once (tensors not present in a file are simply skipped there)
+    FILE* f = fopen(p.c_str(), "rb");
+    if (!f) return set_err(4, "cannot open %s", p.c_str());
+    fseek(f, 0, SEEK_END);
+    long n = ftell(f);
+    fseek(f, 0, SEEK_SET);
+    std::string data((size_t)n, 0);
+    if (fread(&data[0], 1, (size_t)n, f) != (size_t)n) {
+      fclose(f);
+      return set_err(4, "short read on %s", p.c_str());
+    }
+    fclose(f);
+    minijson::ValuePtr v;
+    try {
+      v = minijson::parse(data);
+    } catch (const std::exception& ex) {
+      return set_err(4, "index.json: %s", ex.what());
+    }
+    auto wm = v->get("weight_map");
+    if (!wm || wm->kind != minijson::Value::Obj)
+      return set_err(4, "index.json has no weight_map");
+    std::string dir = p.substr(0, p.find_last_of('/') + 1);
+    std::vector<std::string> files;
+    for (auto& kv : wm->obj) {
+      const std::string& fn = kv.second->str;
+      bool seen = false;
+      for (auto& x : files) seen |= (x == fn);
+      if (!seen) files.push_back(fn);
+    }
+    e->loading_multi = true;
+    e->missing.clear();
+    for (auto& fn : files) {
+      std::string full = dir + fn;
+      int r = load_safetensors_file(e, full.c_str());
+      if (r) {
+        e->loading_multi = false;
+        return r;
+      }
+    }
+    e->loading_multi = false;
+    // final presence check: every required tensor must have been seen
+    if (!e->missing.empty())
+      return set_err(4, "sharded checkpoint is missing tensor %s",
+                     e->missing.begin()->c_str());
+    HIP_TRY(hipDeviceSynchronize());
+    e->weights_ready = true;
+    return 0;
+  }
+  int r = load_safetensors_file(e, p.c_str());
+  if (r) return r;
+  HIP_TRY(hipDeviceSynchronize());
+  e->weights_ready = true;
+  return 0;
+}
+
+static int load_safetensors_file(cake_engine* e, const char* path) {
   HIP_TRY(hipSetDevice(e->device));
   FILE* f = fopen(path, "rb");
   if (!f) return set_err(4, "cannot open %s", path);
@@ -793,27 +880,39 @@ extern "C" int cake_hip_load_safetensors(cake_engine* e, const char* path) {
   const size_t H = c.hidden, I = c.inter, V = c.vocab;
   const size_t hd = c.hd(), Sq = c.sq(), Skv = c.skv();
 
+  // In multi-file mode a tensor may live in another shard: record it as
+  // pending instead of failing; the index loader checks completeness.
   auto need = [&](const std::string& name, StTensor* out_t) -> int {
     auto it = tensors.find(name);
-    if (it == tensors.end()) return set_err(4, "missing tensor %s",
-                                            name.c_str());
+    if (it == tensors.end()) {
+      if (e->loading_multi) {
+        e->missing.insert(name);
+        return -1;  // sentinel: skip this tensor in this file
+      }
+      return set_err(4, "missing tensor %s", name.c_str());
+    }
+    e->missing.erase(name);
     *out_t = it->second;
     return 0;
   };
+#define LOADW(NAME, UPLOAD_EXPR)                                       \
+  do {                                                                 \
+    int _r = need((NAME), &t);                                         \
+    if (_r == -1) break; /* lives in another shard */                  \
+    if (_r) return _r;                                                 \
+    if ((_r = (UPLOAD_EXPR))) return _r;                               \
+  } while (0)
   int r;
   StTensor t;
   if (e->has_embed()) {
-    if ((r = need("model.embed_tokens.weight", &t))) return r;
-    if ((r = upload_weight(e, e->embed, base, t, V * H))) return r;
+    LOADW("model.embed_tokens.weight", upload_weight(e, e->embed, base, t, V * H));
   }
   if (e->has_head()) {
-    if ((r = need("model.norm.weight", &t))) return r;
-    if ((r = upload_weight(e, e->norm_w, base, t, H))) return r;
+    LOADW("model.norm.weight", upload_weight(e, e->norm_w, base, t, H));
     if (!(c.tied && e->has_embed())) {
       std::string name = c.tied ? "model.embed_tokens.weight"
                                 : "lm_head.weight";
-      if ((r = need(name, &t))) return r;
-      if ((r = upload_weight(e, e->lm_head, base, t, V * H))) return r;
+      LOADW(name, upload_weight(e, e->lm_head, base, t, V * H));
     }
   }
   for (int li = e->lo; li < e->hi; ++li) {
@@ -821,77 +920,45 @@ extern "C" int cake_hip_load_safetensors(cake_engine* e, const char* path) {
     char p[128];
     snprintf(p, sizeof p, "model.layers.%d.", li);
     std::string pre(p);
-    if ((r = need(pre + "input_layernorm.weight", &t))) return r;
-    if ((r = upload_weight(e, l.rms1, base, t, H))) return r;
-    if ((r = need(pre + "post_attention_layernorm.weight", &t))) return r;
-    if ((r = upload_weight(e, l.rms2, base, t, H))) return r;
+    LOADW(pre + "input_layernorm.weight", upload_weight(e, l.rms1, base, t, H));
+    LOADW(pre + "post_attention_layernorm.weight", upload_weight(e, l.rms2, base, t, H));
     if (c.fp8) {
       // fp8 weights + blockwise scale_inv (fp8.rs:42-64); fused qkv and
       // gate_up concatenate both the byte tensors and the scale rows
       const size_t Hb = H / 128, Sqb = Sq / 128, Skvb = Skv / 128,
                    Ib = I / 128;
-      if ((r = need(pre + "self_attn.q_proj.weight", &t))) return r;
-      if ((r = upload_weight_u8(e, l.wqkv8, base, t, Sq * H))) return r;
-      if ((r = need(pre + "self_attn.k_proj.weight", &t))) return r;
-      if ((r = upload_weight_u8(e, l.wqkv8 + Sq * H, base, t, Skv * H)))
-        return r;
-      if ((r = need(pre + "self_attn.v_proj.weight", &t))) return r;
-      if ((r = upload_weight_u8(e, l.wqkv8 + (Sq + Skv) * H, base, t,
-                                Skv * H)))
-        return r;
-      if ((r = need(pre + "self_attn.q_proj.weight_scale_inv", &t))) return r;
-      if ((r = upload_scale(e, l.sqkv, base, t, Sqb * Hb))) return r;
-      if ((r = need(pre + "self_attn.k_proj.weight_scale_inv", &t))) return r;
-      if ((r = upload_scale(e, l.sqkv + Sqb * Hb, base, t, Skvb * Hb)))
-        return r;
-      if ((r = need(pre + "self_attn.v_proj.weight_scale_inv", &t))) return r;
-      if ((r = upload_scale(e, l.sqkv + (Sqb + Skvb) * Hb, base, t,
-                            Skvb * Hb)))
-        return r;
-      if ((r = need(pre + "self_attn.o_proj.weight", &t))) return r;
-      if ((r = upload_weight_u8(e, l.wo8, base, t, H * Sq))) return r;
-      if ((r = need(pre + "self_attn.o_proj.weight_scale_inv", &t))) return r;
-      if ((r = upload_scale(e, l.so8, base, t, Hb * Sqb))) return r;
-      if ((r = need(pre + "mlp.gate_proj.weight", &t))) return r;
-      if ((r = upload_weight_u8(e, l.wgu8, base, t, I * H))) return r;
-      if ((r = need(pre + "mlp.up_proj.weight", &t))) return r;
-      if ((r = upload_weight_u8(e, l.wgu8 + I * H, base, t, I * H))) return r;
-      if ((r = need(pre + "mlp.gate_proj.weight_scale_inv", &t))) return r;
-      if ((r = upload_scale(e, l.sgu, base, t, Ib * Hb))) return r;
-      if ((r = need(pre + "mlp.up_proj.weight_scale_inv", &t))) return r;
-      if ((r = upload_scale(e, l.sgu + Ib * Hb, base, t, Ib * Hb))) return r;
-      if ((r = need(pre + "mlp.down_proj.weight", &t))) return r;
-      if ((r = upload_weight_u8(e, l.wdown8, base, t, H * I))) return r;
-      if ((r = need(pre + "mlp.down_proj.weight_scale_inv", &t))) return r;
-      if ((r = upload_scale(e, l.sdown, base, t, Hb * Ib))) return r;
+      LOADW(pre + "self_attn.q_proj.weight", upload_weight_u8(e, l.wqkv8, base, t, Sq * H));
+      LOADW(pre + "self_attn.k_proj.weight", upload_weight_u8(e, l.wqkv8 + Sq * H, base, t, Skv * H));
+      LOADW(pre + "self_attn.v_proj.weight", upload_weight_u8(e, l.wqkv8 + (Sq + Skv) * H, base, t, Skv * H));
+      LOADW(pre + "self_attn.q_proj.weight_scale_inv", upload_scale(e, l.sqkv, base, t, Sqb * Hb));
+      LOADW(pre + "self_attn.k_proj.weight_scale_inv", upload_scale(e, l.sqkv + Sqb * Hb, base, t, Skvb * Hb));
+      LOADW(pre + "self_attn.v_proj.weight_scale_inv", upload_scale(e, l.sqkv + (Sqb + Skvb) * Hb, base, t, Skvb * Hb));
+      LOADW(pre + "self_attn.o_proj.weight", upload_weight_u8(e, l.wo8, base, t, H * Sq));
+      LOADW(pre + "self_attn.o_proj.weight_scale_inv", upload_scale(e, l.so8, base, t, Hb * Sqb));
+      LOADW(pre + "mlp.gate_proj.weight", upload_weight_u8(e, l.wgu8, base, t, I * H));
+      LOADW(pre + "mlp.up_proj.weight", upload_weight_u8(e, l.wgu8 + I * H, base, t, I * H));
+      LOADW(pre + "mlp.gate_proj.weight_scale_inv", upload_scale(e, l.sgu, base, t, Ib * Hb));
+      LOADW(pre + "mlp.up_proj.weight_scale_inv", upload_scale(e, l.sgu + Ib * Hb, base, t, Ib * Hb));
+      LOADW(pre + "mlp.down_proj.weight", upload_weight_u8(e, l.wdown8, base, t, H * I));
+      LOADW(pre + "mlp.down_proj.weight_scale_inv", upload_scale(e, l.sdown, base, t, Hb * Ib));
     } else {
     // fused qkv: upload q,k,v at row offsets (attention.rs:109-114)
-    if ((r = need(pre + "self_attn.q_proj.weight", &t))) return r;
-    if ((r = upload_weight(e, l.wqkv, base, t, Sq * H))) return r;
-    if ((r = need(pre + "self_attn.k_proj.weight", &t))) return r;
-    if ((r = upload_weight(e, l.wqkv + Sq * H, base, t, Skv * H))) return r;
-    if ((r = need(pre + "self_attn.v_proj.weight", &t))) return r;
-    if ((r = upload_weight(e, l.wqkv + (Sq + Skv) * H, base, t, Skv * H)))
-      return r;
-    if ((r = need(pre + "self_attn.o_proj.weight", &t))) return r;
-    if ((r = upload_weight(e, l.wo, base, t, H * Sq))) return r;
+    LOADW(pre + "self_attn.q_proj.weight", upload_weight(e, l.wqkv, base, t, Sq * H));
+    LOADW(pre + "self_attn.k_proj.weight", upload_weight(e, l.wqkv + Sq * H, base, t, Skv * H));
+    LOADW(pre + "self_attn.v_proj.weight", upload_weight(e, l.wqkv + (Sq + Skv) * H, base, t, Skv * H));
+    LOADW(pre + "self_attn.o_proj.weight", upload_weight(e, l.wo, base, t, H * Sq));
     // fused gate_up (mlp.rs:38-46)
-    if ((r = need(pre + "mlp.gate_proj.weight", &t))) return r;
-    if ((r = upload_weight(e, l.wgu, base, t, I * H))) return r;
-    if ((r = need(pre + "mlp.up_proj.weight", &t))) return r;
-    if ((r = upload_weight(e, l.wgu + I * H, base, t, I * H))) return r;
-    if ((r = need(pre + "mlp.down_proj.weight", &t))) return r;
-    if ((r = upload_weight(e, l.wdown, base, t, H * I))) return r;
+    LOADW(pre + "mlp.gate_proj.weight", upload_weight(e, l.wgu, base, t, I * H));
+    LOADW(pre + "mlp.up_proj.weight", upload_weight(e, l.wgu + I * H, base, t, I * H));
+    LOADW(pre + "mlp.down_proj.weight", upload_weight(e, l.wdown, base, t, H * I));
     }
     if (c.qk_norm) {
-      if ((r = need(pre + "self_attn.q_norm.weight", &t))) return r;
-      if ((r = upload_weight(e, l.qnorm, base, t, hd))) return r;
-      if ((r = need(pre + "self_attn.k_norm.weight", &t))) return r;
-      if ((r = upload_weight(e, l.knorm, base, t, hd))) return r;
+      LOADW(pre + "self_attn.q_norm.weight", upload_weight(e, l.qnorm, base, t, hd));
+      LOADW(pre + "self_attn.k_norm.weight", upload_weight(e, l.knorm, base, t, hd));
     }
   }
+#undef LOADW
   HIP_TRY(hipDeviceSynchronize());
-  e->weights_ready = true;
   return 0;
 }
 
