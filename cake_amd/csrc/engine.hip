@@ -257,6 +257,7 @@ struct cake_engine {
   // multi-file (sharded) checkpoint loading state
   bool loading_multi = false;
   std::set<std::string> missing;
+  std::set<std::string> loaded;
 };
 
 static int dev_alloc(void** p, size_t bytes) {
@@ -814,6 +815,7 @@ extern "C" int cake_hip_load_safetensors(cake_engine* e, const char* path) {
     }
     e->loading_multi = true;
     e->missing.clear();
+    e->loaded.clear();
     for (auto& fn : files) {
       std::string full = dir + fn;
       int r = load_safetensors_file(e, full.c_str());
@@ -883,15 +885,20 @@ static int load_safetensors_file(cake_engine* e, const char* path) {
   // In multi-file mode a tensor may live in another shard: record it as
   // pending instead of failing; the index loader checks completeness.
   auto need = [&](const std::string& name, StTensor* out_t) -> int {
+    if (e->loading_multi && e->loaded.count(name))
+      return -1;  // already uploaded from an earlier shard
     auto it = tensors.find(name);
     if (it == tensors.end()) {
       if (e->loading_multi) {
         e->missing.insert(name);
-        return -1;  // sentinel: skip this tensor in this file
+        return -1;  // sentinel: maybe in a later shard
       }
       return set_err(4, "missing tensor %s", name.c_str());
     }
-    e->missing.erase(name);
+    if (e->loading_multi) {
+      e->loaded.insert(name);
+      e->missing.erase(name);
+    }
     *out_t = it->second;
     return 0;
   };
