@@ -194,7 +194,10 @@ def main():
     if args.tokenizer:
         from tokenizers import Tokenizer
         tok = Tokenizer.from_file(args.tokenizer)
-    app = create_app(eng, model_name=args.model, tokenizer=tok)
+    eos = cfg.get("eos_token_id")
+    eos_ids = ([eos] if isinstance(eos, int) else (eos or []))
+    app = create_app(eng, model_name=args.model, tokenizer=tok,
+                     eos_ids=eos_ids)
 
     import uvicorn
     uvicorn.run(app, host=args.host, port=args.port)
