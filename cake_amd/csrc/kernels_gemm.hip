@@ -149,6 +149,155 @@ __global__ __launch_bounds__(256) void k_gemm_bf16(
 // overwritten two tiles after its last read (>= 2 barriers apart).
 // Requires K % 64 == 0; M/N tails handled by clamped loads + guarded
 // stores.  Dispatched only when the grid has >= GEMM256_MIN_BLOCKS tiles.
+
+// ---------------------------------------------------------------------------
+// 128x256 GEMM with TWO-TILE lookahead (3 staging buffers, 144 KB LDS).
+// The 256^2 kernel's single-tile lookahead (~1 phase of cover) measured
+// 38% SQ_WAIT_ANY; staging tile t+2 while computing tile t gives ~2 phases
+// of flight per piece.  8 waves as 2(M) x 4(N), per-wave 64x64 output
+// (16 acc f32x4).  Pieces per tile: A (128x64 = 16 KB, 2 glds/wave),
+// B (256x64 = 32 KB, 4 glds/wave); phase (t,ks0) stages A(t+2), phase
+// (t,ks1) stages B(t+2).  FIFO counts: wait vmcnt(8) at ks0 (A(t+1) +
+// B(t+1) + A(t+2) outstanding), vmcnt(12) at ks1.
+// ---------------------------------------------------------------------------
+template <int EPI>
+__global__ __launch_bounds__(512) void k_gemm_128x256(
+    const u16* __restrict__ A, const u16* __restrict__ W, u16* __restrict__ C,
+    const u16* __restrict__ res, int M, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  auto apiece = [&](int buf) -> u16* {
+    return reinterpret_cast<u16*>(smem + buf * 16384);
+  };
+  auto bpiece = [&](int buf) -> u16* {
+    return reinterpret_cast<u16*>(smem + 3 * 16384 + buf * 32768);
+  };
+  int mt, nt;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    const int orig = blockIdx.x + gridDim.x * blockIdx.y;
+    const int q = nwg / 8, rr = nwg % 8;
+    const int xcd = orig % 8, idx = orig / 8;
+    const int wgid =
+        (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+    mt = wgid % gridDim.x;
+    nt = wgid / gridDim.x;
+  }
+  const int m0 = mt * 128, n0 = nt * 256;
+  const int t = threadIdx.x;
+  const int w = t / WAVE, lane = t % WAVE;
+  const int wm = w >> 2, wn = w & 3;  // 2 x 4 waves, 64x64 each
+
+  auto stage_a = [&](int tau) {
+    u16* dst = apiece(tau % 3);
+    const int kt = tau * 64;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int rloc = (w * 2 + j) * 8 + lane / 8;  // 0..127
+      const int grow = min(m0 + rloc, M - 1);
+      const int u = (lane % 8) ^ (rloc & 7);
+      const u16* src = A + (size_t)grow * K + kt + u * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(uintptr_t)src,
+          (__attribute__((address_space(3))) void*)(uintptr_t)(
+              dst + (size_t)(w * 2 + j) * 8 * 64),
+          16, 0, 0);
+    }
+  };
+  auto stage_b = [&](int tau) {
+    u16* dst = bpiece(tau % 3);
+    const int kt = tau * 64;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int rloc = (w * 4 + j) * 8 + lane / 8;  // 0..255
+      const int grow = min(n0 + rloc, N - 1);
+      const int u = (lane % 8) ^ (rloc & 7);
+      const u16* src = W + (size_t)grow * K + kt + u * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(uintptr_t)src,
+          (__attribute__((address_space(3))) void*)(uintptr_t)(
+              dst + (size_t)(w * 4 + j) * 8 * 64),
+          16, 0, 0);
+    }
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int NT = K / 64;
+  stage_a(0);
+  stage_b(0);
+  if (NT > 1) {
+    stage_a(1);
+    stage_b(1);
+  }
+  for (int tau = 0; tau < NT; ++tau) {
+    const u16* ap = apiece(tau % 3);
+    const u16* bp = bpiece(tau % 3);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      if (ks == 0) {
+        if (tau + 2 < NT) {
+          stage_a(tau + 2);
+          asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+        } else {
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+      } else {
+        if (tau + 2 < NT) {
+          stage_b(tau + 2);
+          asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+        }
+        // tail tiles: everything needed was already drained at ks0
+      }
+      __builtin_amdgcn_s_barrier();
+      bf16x8 af[4], bf[4];
+      const int u = ks * 4 + (lane / 16);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        const int r = wm * 64 + mi * 16 + (lane % 16);
+        af[mi] = *reinterpret_cast<const bf16x8*>(
+            ap + (size_t)r * 64 + (size_t)(u ^ (r & 7)) * 8);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int r = wn * 64 + ni * 16 + (lane % 16);
+        bf[ni] = *reinterpret_cast<const bf16x8*>(
+            bp + (size_t)r * 64 + (size_t)(u ^ (r & 7)) * 8);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wm * 64 + mi * 16 + (lane / 16) * 4 + r;
+        const int col = n0 + wn * 64 + ni * 16 + (lane % 16);
+        if (row < M && col < N) {
+          float v = acc[mi][ni][r];
+          if (EPI == 1) v += b2f(res[(size_t)row * N + col]);
+          C[(size_t)row * N + col] = f2b(v);
+        }
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 template <int EPI>
 __global__ __launch_bounds__(512) void k_gemm_256(
@@ -274,14 +423,36 @@ __global__ __launch_bounds__(512) void k_gemm_256(
 
 void launch_gemm(const u16* A, const u16* W, u16* C, const u16* res, int M,
                  int N, int K, int epi, hipStream_t s) {
-  // 256^2 counted-vmcnt kernel when the grid still fills the chip at one
-  // 8-wave block per CU (CAKE_GEMM256=0 disables for A/B)
-  static const int use256 = [] {
-    const char* v = getenv("CAKE_GEMM256");
-    return v ? atoi(v) : 1;
+  // variant select: 0 = 128^2 baseline, 1 = 256^2 counted-vmcnt,
+  // 2 = 128x256 3-buffer two-tile lookahead (CAKE_GEMM_VAR overrides)
+  static const int var = [] {
+    const char* v = getenv("CAKE_GEMM_VAR");
+    if (v) return atoi(v);
+    const char* o = getenv("CAKE_GEMM256");  // legacy knob
+    if (o && atoi(o) == 0) return 0;
+    return 1;
   }();
+  const int mt128 = (M + 127) / 128, nt256 = (N + 255) / 256;
+  if (var == 2 && (long)mt128 * nt256 >= 200 && K % 64 == 0) {
+    static bool attr2 = false;
+    if (!attr2) {
+      hipFuncSetAttribute((const void*)&k_gemm_128x256<0>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, 147456);
+      hipFuncSetAttribute((const void*)&k_gemm_128x256<1>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, 147456);
+      attr2 = true;
+    }
+    dim3 grid(mt128, nt256);
+    if (epi == 0)
+      hipLaunchKernelGGL(k_gemm_128x256<0>, grid, dim3(512), 147456, s, A, W,
+                         C, res, M, N, K);
+    else
+      hipLaunchKernelGGL(k_gemm_128x256<1>, grid, dim3(512), 147456, s, A, W,
+                         C, res, M, N, K);
+    return;
+  }
   const int mt = (M + 255) / 256, nt = (N + 255) / 256;
-  if (use256 && (long)mt * nt >= 200) {
+  if (var >= 1 && (long)mt * nt >= 200) {
     static bool attr_set = false;
     if (!attr_set) {
       hipFuncSetAttribute((const void*)&k_gemm_256<0>,
