@@ -235,6 +235,9 @@ struct cake_engine {
   int* pidx = nullptr;
   float* attn_ws = nullptr;
   u32* attn_cnt = nullptr;
+  float* gemv_ws = nullptr;   // split-K GEMV partials [N][2]
+  u32* gemv_cnt = nullptr;    // split-K arrival counters (epoch-free)
+  int splitk = 0;             // CAKE_GEMV_SPLITK (0 = off)
   int* dev_pos = nullptr;
   int* dev_step = nullptr;
   u32* dev_tok = nullptr;
@@ -367,6 +370,9 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
     if (c.fp8)
       launch_gemv_fp8(l.wo8, l.so8, e->attn_out, e->x, e->x, nullptr, 0.f,
                       H, Sq, 1, e->stream);
+    else if (e->splitk == 2 && Sq % 16 == 0)
+      launch_gemv_res_splitk(l.wo, e->attn_out, e->x, e->x, e->gemv_ws,
+                             e->gemv_cnt, H, Sq, e->stream);
     else
       launch_gemv(l.wo, e->attn_out, e->x, e->x, nullptr, 0.f, H, Sq, 1,
                   e->stream);
@@ -387,6 +393,9 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
     if (c.fp8)
       launch_gemv_fp8(l.wdown8, l.sdown, e->act, e->x, e->x, nullptr, 0.f,
                       H, I, 1, e->stream);
+    else if (e->splitk == 2 && I % 16 == 0)
+      launch_gemv_res_splitk(l.wdown, e->act, e->x, e->x, e->gemv_ws,
+                             e->gemv_cnt, H, I, e->stream);
     else
       launch_gemv(l.wdown, e->act, e->x, e->x, nullptr, 0.f, H, I, 1,
                   e->stream);
@@ -641,6 +650,14 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
   ALLOC(e->attn_ws, float, (size_t)c.nh * 64 * (hd + 2));  // nchunk <= 64
   ALLOC(e->attn_cnt, u32, c.nh);
   HIP_TRY(hipMemset(e->attn_cnt, 0, sizeof(u32) * c.nh));
+  if (const char* sk = getenv("CAKE_GEMV_SPLITK"))
+    e->splitk = atoi(sk);
+  {
+    const size_t mx = (size_t)std::max(H, I);
+    ALLOC(e->gemv_ws, float, mx * 2);
+    ALLOC(e->gemv_cnt, u32, (mx + 1) / 2);
+    HIP_TRY(hipMemset(e->gemv_cnt, 0, sizeof(u32) * ((mx + 1) / 2)));
+  }
   ALLOC(e->dev_pos, int, 1);
   ALLOC(e->dev_step, int, 1);
   ALLOC(e->dev_tok, u32, 1);

@@ -26,6 +26,9 @@ void launch_gemv(const u16* W, const u16* x, void* out, const u16* res,
                  hipStream_t s);
 void launch_gemv_gateup(const u16* W, const u16* x, u16* out, const u16* nw,
                         float eps, int I, int K, int rows, hipStream_t s);
+void launch_gemv_res_splitk(const u16* W, const u16* x, u16* out,
+                            const u16* res, float* ws, u32* cnt, int N,
+                            int K, hipStream_t s);
 void launch_gemv_fp8(const unsigned char* W, const float* sc, const u16* x,
                      void* out, const u16* res, const u16* nw, float eps,
                      int N, int K, int epi, hipStream_t s);

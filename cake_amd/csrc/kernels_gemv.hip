@@ -578,6 +578,112 @@ __global__ void k_dequant_fp8(const unsigned char* __restrict__ W,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Split-K GEMV for the residual projections (o_proj, down_proj): the
+// full-K kernels are memory-latency-bound per block (PMC: 50-88%
+// SQ_WAIT_ANY), so halving each block's serial K-chain and doubling the
+// grid buys overlap.  Partials go through write-through sc1 stores + an
+// epoch-free arrival counter (the attention kernel's proven Guideline 16
+// R1 variant); the last-arriving slice adds the residual and stores bf16.
+// ws: [N][KS] f32; cnt: [N/ROWS] u32 (monotonic, modulo-KS election).
+// ---------------------------------------------------------------------------
+template <int ROWS, int KB, int KS>
+__global__ __launch_bounds__(256) void k_gemv_splitk_res(
+    const u16* __restrict__ W, const u16* __restrict__ x,
+    u16* __restrict__ out, const u16* __restrict__ res,
+    float* __restrict__ ws, u32* __restrict__ cnt, int N, int K) {
+  const int t = threadIdx.x;
+  const int row0 = blockIdx.x * ROWS;
+  const int ksl = blockIdx.y;
+  const int kpart = K / KS;                 // multiple of 8 by dispatch
+  const int kbase = ksl * kpart;
+  const int wid = t / WAVE, lane = t % WAVE;
+  __shared__ float red[ROWS > 4 ? ROWS : 4][4];
+
+  short8 wpre[ROWS][KB];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r)
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (row0 + r < N && k0 < kpart)
+        wpre[r][i] = ntload8(W + (size_t)(row0 + r) * K + kbase + k0);
+    }
+  float xr[KB * 8];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    if (k0 < kpart) {
+      short8 xv = *reinterpret_cast<const short8*>(x + kbase + k0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xv[j]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = 0.f;
+    }
+  }
+  float acc[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) acc[r] = 0.f;
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r)
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (row0 + r < N && k0 < kpart) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc[r] = fmaf(b2f((u16)wpre[r][i][j]), xr[i * 8 + j], acc[r]);
+      }
+    }
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r) {
+    float v = wave_sum(acc[r]);
+    if (lane == 0) red[r][wid] = v;
+  }
+  __syncthreads();
+  // publish partials (sc1 4-B agent stores) and elect the combiner
+  if (t < ROWS && row0 + t < N) {
+    const float v = red[t][0] + red[t][1] + red[t][2] + red[t][3];
+    __hip_atomic_store(&ws[(size_t)(row0 + t) * KS + ksl], v,
+                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (t == 0) {
+    u32 v = __hip_atomic_fetch_add(&cnt[blockIdx.x], 1u, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+    red[0][0] = (v % (u32)KS == (u32)(KS - 1)) ? 1.f : 0.f;
+  }
+  __syncthreads();
+  if (red[0][0] == 0.f) return;
+  if (t < ROWS && row0 + t < N) {
+    float v = 0.f;
+#pragma unroll
+    for (int k = 0; k < KS; ++k)
+      v += __hip_atomic_load(&ws[(size_t)(row0 + t) * KS + k],
+                             __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    out[row0 + t] = f2b(v + b2f(res[row0 + t]));
+  }
+}
+
+void launch_gemv_res_splitk(const u16* W, const u16* x, u16* out,
+                            const u16* res, float* ws, u32* cnt, int N,
+                            int K, hipStream_t s) {
+  // K/2 per slice; pick KB for the half-K
+  const int kpart = K / 2;
+  dim3 grid((N + 1) / 2, 2);
+#define SPK(KB)                                                             \
+  hipLaunchKernelGGL((k_gemv_splitk_res<2, KB, 2>), grid, dim3(256), 0, s,  \
+                     W, x, out, res, ws, cnt, N, K)
+  if (kpart <= 2048) SPK(1);
+  else if (kpart <= 4096) SPK(2);
+  else if (kpart <= 8192) SPK(4);
+  else SPK(8);
+#undef SPK
+}
+
 template <int ROWS, int EPI>
 static void gemv_dispatch_kb(const u16* W, const u16* x, void* out,
                              const u16* res, const u16* nw, float eps, int N,
