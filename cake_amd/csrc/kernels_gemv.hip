@@ -24,7 +24,28 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
   const int wid = t / WAVE, lane = t % WAVE;
   __shared__ float red[ROWS > 4 ? ROWS : 4][4];
 
-  // For short K, issue ALL weight loads first — they stay in flight across
+  // Load issue order = consumer order (guide T20 follow-on: a wait
+  // belongs at the first consumer, and hipcc's guarded blocks degrade
+  // counted waits to vmcnt(0) drains — so the group whose consumer comes
+  // FIRST must issue first).  x feeds the sumsq immediately; the norm
+  // weights are consumed after the reduction barrier; the row weights
+  // last, in the FMA phase.  All three groups are in flight together, so
+  // a drain before sumsq costs one memory latency, not three.
+  short8 xpre[KB];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    if (k0 < K) xpre[i] = *reinterpret_cast<const short8*>(x + k0);
+  }
+  short8 nwpre[NORM ? KB : 1];
+  if (NORM) {
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) nwpre[i] = *reinterpret_cast<const short8*>(nw + k0);
+    }
+  }
+  // For short K, issue ALL weight loads too — they stay in flight across
   // the x/norm phase (plain VGPR loads survive s_barrier; guide §5
   // pipelining note), hiding the norm reduction latency entirely.
   short8 wpre[KB <= 2 ? ROWS : 1][KB <= 2 ? KB : 1];
@@ -39,15 +60,14 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
       }
   }
 
-  // phase 1: x -> registers (f32), optionally fused rms_norm
+  // phase 1: x -> f32, optionally fused rms_norm
   float xr[KB * 8];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
     const int k0 = i * 2048 + t * 8;
     if (k0 < K) {
-      short8 xv = *reinterpret_cast<const short8*>(x + k0);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xv[j]);
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xpre[i][j]);
     } else {
 #pragma unroll
       for (int j = 0; j < 8; ++j) xr[i * 8 + j] = 0.f;
@@ -68,11 +88,11 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
     for (int i = 0; i < KB; ++i) {
       const int k0 = i * 2048 + t * 8;
       if (k0 < K) {
-        short8 wv = *reinterpret_cast<const short8*>(nw + k0);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           // re-quantize so fused == rmsnorm-kernel-then-gemv bit-exactly
-          xr[i * 8 + j] = b2f(f2b(xr[i * 8 + j] * scale * b2f((u16)wv[j])));
+          xr[i * 8 + j] =
+              b2f(f2b(xr[i * 8 + j] * scale * b2f((u16)nwpre[i][j])));
       }
     }
   }
@@ -202,14 +222,29 @@ __global__ __launch_bounds__(256) void k_gemv_gateup(
   const int wid = t / WAVE, lane = t % WAVE;
   __shared__ float redg[8][4], redu[8][4];
 
+  // load issue order = consumer order (see k_gemv_reg): x first (feeds
+  // sumsq), then norm weights (consumed after the barrier)
+  short8 xpre[KB];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    if (k0 < K) xpre[i] = *reinterpret_cast<const short8*>(x + k0);
+  }
+  short8 nwpre[NORM ? KB : 1];
+  if (NORM) {
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) nwpre[i] = *reinterpret_cast<const short8*>(nw + k0);
+    }
+  }
   float xr[KB * 8];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
     const int k0 = i * 2048 + t * 8;
     if (k0 < K) {
-      short8 xv = *reinterpret_cast<const short8*>(x + k0);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xv[j]);
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xpre[i][j]);
     } else {
 #pragma unroll
       for (int j = 0; j < 8; ++j) xr[i * 8 + j] = 0.f;
@@ -230,10 +265,10 @@ __global__ __launch_bounds__(256) void k_gemv_gateup(
     for (int i = 0; i < KB; ++i) {
       const int k0 = i * 2048 + t * 8;
       if (k0 < K) {
-        short8 wv = *reinterpret_cast<const short8*>(nw + k0);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          xr[i * 8 + j] = b2f(f2b(xr[i * 8 + j] * scale * b2f((u16)wv[j])));
+          xr[i * 8 + j] =
+              b2f(f2b(xr[i * 8 + j] * scale * b2f((u16)nwpre[i][j])));
       }
     }
   }
@@ -305,7 +340,10 @@ __global__ __launch_bounds__(256) void k_gemv_fp8(
         wpre[r][i] = ntload16b(W + (size_t)(row0 + r) * K + k0);
     }
 
-  // x -> registers (f32), optionally fused rms_norm
+  // x -> registers (f32), optionally fused rms_norm.  (The bf16 kernels'
+  // early nw/x prefetch is a measured NEGATIVE here: fp8 runs at lower
+  // occupancy and the extra live short8 registers cost more than the
+  // hidden latency buys — A/B'd -5% on qwen3-32b-fp8 decode.)
   float xr[KB * 16];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
@@ -468,6 +506,8 @@ __global__ __launch_bounds__(256) void k_gemv_gateup_fp8(
   const int wid = t / WAVE, lane = t % WAVE;
   __shared__ float redg[8][4], redu[8][4];
 
+  // x -> registers; early nw/x prefetch is a measured NEGATIVE on the fp8
+  // path (occupancy, see k_gemv_fp8 note)
   float xr[KB * 16];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
