@@ -1130,25 +1130,45 @@ extern "C" int cake_hip_decode(cake_engine* e, int steps,
   if (e->host_pos + steps > e->max_seq)
     return set_err(5, "decode exceeds max_seq");
   HIP_TRY(hipMemsetAsync(e->dev_step, 0, 4, e->stream));
-  // pick the split-KV chunk count for the expected context length (fixed
-  // once a graph is captured; CAKE_NCHUNK overrides)
-  if (!getenv("CAKE_NCHUNK") && !e->graph)
-    e->nchunk = e->host_pos >= 1024 ? 16 : 8;
+  // Split-KV chunk count for decode attention, by context length (measured
+  // optima on MI355X, r01 sweep: ctx<1024 -> 8, 2048 -> 24, >=3960 -> 32).
+  // The graph bakes the grid in, so when the context grows across a
+  // threshold the graph is dropped and lazily re-captured with the new
+  // chunk count (capture costs ~1 step, amortized over thousands).
+  const bool nchunk_fixed = getenv("CAKE_NCHUNK") != nullptr;
+  auto want_nchunk = [](int pos) {
+    return pos < 1024 ? 8 : pos < 3072 ? 24 : 32;
+  };
   bool graph_ok = e->use_graph() && e->world == 1 && !e->st.on;
-  if (graph_ok && !e->graph) {
-    // capture one decode step (device-side token/pos/ring make it replayable)
-    hipGraph_t g;
-    HIP_TRY(hipStreamBeginCapture(e->stream, hipStreamCaptureModeGlobal));
-    int r = enqueue_decode_step(e);
-    if (r) {
-      hipStreamEndCapture(e->stream, &g);
-      return r;
+  if (!nchunk_fixed) {
+    const int want = want_nchunk(e->host_pos);
+    if (want != e->nchunk) {
+      if (e->graph) {
+        HIP_TRY(hipGraphExecDestroy(e->graph));
+        e->graph = nullptr;
+      }
+      e->nchunk = want;
+      // the epoch-free arrival counters advance by nchunk per step, so a
+      // chunk-count change invalidates their modulo election: re-zero
+      // (stream is idle here — the previous call ended in a sync)
+      HIP_TRY(hipMemset(e->attn_cnt, 0, sizeof(u32) * e->c.nh));
     }
-    HIP_TRY(hipStreamEndCapture(e->stream, &g));
-    HIP_TRY(hipGraphInstantiate(&e->graph, g, nullptr, nullptr, 0));
-    HIP_TRY(hipGraphDestroy(g));
   }
   for (int s = 0; s < steps; ++s) {
+    if (graph_ok && !e->graph) {
+      // capture one decode step (device-side token/pos/ring make it
+      // replayable)
+      hipGraph_t g;
+      HIP_TRY(hipStreamBeginCapture(e->stream, hipStreamCaptureModeGlobal));
+      int r = enqueue_decode_step(e);
+      if (r) {
+        hipStreamEndCapture(e->stream, &g);
+        return r;
+      }
+      HIP_TRY(hipStreamEndCapture(e->stream, &g));
+      HIP_TRY(hipGraphInstantiate(&e->graph, g, nullptr, nullptr, 0));
+      HIP_TRY(hipGraphDestroy(g));
+    }
     if (graph_ok) {
       HIP_TRY(hipGraphLaunch(e->graph, e->stream));
     } else {
@@ -1156,8 +1176,22 @@ extern "C" int cake_hip_decode(cake_engine* e, int steps,
       if (r) return r;
     }
     e->host_pos += 1;
-    // bound in-flight work: sync every 64 steps to keep the queue shallow
-    if ((s & 63) == 63) HIP_TRY(hipStreamSynchronize(e->stream));
+    // bound in-flight work: sync every 64 steps to keep the queue shallow;
+    // same cadence re-checks the chunk-count threshold as context grows
+    if ((s & 63) == 63) {
+      HIP_TRY(hipStreamSynchronize(e->stream));
+      if (!nchunk_fixed) {
+        const int want = want_nchunk(e->host_pos);
+        if (want != e->nchunk) {
+          if (e->graph) {
+            HIP_TRY(hipGraphExecDestroy(e->graph));
+            e->graph = nullptr;
+          }
+          e->nchunk = want;
+          HIP_TRY(hipMemset(e->attn_cnt, 0, sizeof(u32) * e->c.nh));
+        }
+      }
+    }
   }
   HIP_TRY(hipStreamSynchronize(e->stream));
   stats_flush(e);

@@ -421,6 +421,49 @@ def test_long_context_decode_parity():
         del _os.environ["CAKE_NCHUNK"]
 
 
+def test_adaptive_nchunk_recapture():
+    """Crossing the 1024-context threshold mid-decode drops and re-captures
+    the decode graph with a larger split-KV chunk count (engine.hip
+    want_nchunk) and re-zeros the arrival counters.  Decode across the
+    boundary must stay consistent with the uncached forward and the
+    oracle."""
+    import os
+    import tempfile
+    cfg_json = dict(
+        model_type="llama", hidden_size=256, intermediate_size=512,
+        vocab_size=512, num_hidden_layers=2, num_attention_heads=2,
+        num_key_value_heads=1, head_dim=128, rms_norm_eps=1e-5,
+        rope_theta=500000.0, max_position_embeddings=4096,
+        tie_word_embeddings=False)
+    cfg = Config.from_json(cfg_json)
+    w = random_weights(cfg, seed=177)
+    oracle = quantized_oracle(cfg, w)
+    with tempfile.TemporaryDirectory() as td:
+        st = os.path.join(td, "m.safetensors")
+        weights_to_safetensors(w, cfg, st)
+        eng = cake_amd.Engine(json.dumps(cfg_json), max_seq=2048,
+                              max_batch_tokens=1024)
+        eng.load_safetensors(st)
+        try:
+            rng = np.random.default_rng(31)
+            prompt = rng.integers(0, cfg.vocab_size,
+                                  size=990).astype(np.uint32)
+            first = eng.prefill(prompt)
+            # 140 graph-replayed steps: host_pos crosses 1024 at the s=63
+            # sync point -> re-capture with nchunk 24
+            toks = eng.decode(140)
+            seq = np.concatenate([prompt, [first], toks[:-1]]).astype(
+                np.uint32)
+            eng.reset()
+            _, lg = eng.prefill(seq, want_logits=True)
+            assert int(np.argmax(lg)) == int(toks[-1])
+            oracle.reset()
+            ref = oracle.forward(seq[None, :].astype(np.int64), 0)[0]
+            assert rel_err(lg, ref) < 2e-2
+        finally:
+            eng.close()
+
+
 def test_max_seq_guard():
     cfg_json = dict(
         model_type="llama", hidden_size=64, intermediate_size=128,
