@@ -180,20 +180,28 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
   }
   float m = -INFINITY, l = 0.f, o0 = 0.f, o1 = 0.f;
   const int TILE = 128;
+  // Loads below are UNGUARDED (addresses clamped into the cache, results
+  // zeroed/-INF'd by weight instead): a load inside an `if (p < end)`
+  // block costs hipcc a vmcnt(0) drain per unrolled iteration (the .s
+  // showed 24 of them = zero memory-level parallelism, 0.5 TB/s at 8k
+  // context); straight-line loads issue back-to-back and wait once.
+  // Clamped addresses stay inside this kv-head's cache rows, and qa[] is
+  // zero beyond hd, so out-of-range lanes contribute exact zeros.
+  const int doff = min(dgrp * 8, hd - 8);
+  const int eoff = min(e0, hd - 2);
   for (int sub0 = start; sub0 < end; sub0 += TILE) {
     // --- phase A: scores for [sub0, sub0+TILE) --------------------------
     // 256 threads = 16 positions per pass (16 lanes per position, each
     // loading 16 B of the K row)
-#pragma unroll 4
+#pragma unroll
     for (int pass = 0; pass < TILE / 16; ++pass) {
       const int p = sub0 + pass * 16 + (t >> 4);
+      const int pc = max(0, min(p, end - 1));
+      const u16* kr = kbase + (size_t)pc * hd + doff;
+      short8 kv8 = *reinterpret_cast<const short8*>(kr);
       float d = 0.f;
-      if (p < end && dgrp * 8 < hd) {
-        const u16* kr = kbase + (size_t)p * hd + dgrp * 8;
-        short8 kv8 = *reinterpret_cast<const short8*>(kr);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) d = fmaf(b2f((u16)kv8[j]), qa[j], d);
-      }
+      for (int j = 0; j < 8; ++j) d = fmaf(b2f((u16)kv8[j]), qa[j], d);
       // reduce across the 16 lanes of this position
 #pragma unroll
       for (int off = 8; off >= 1; off >>= 1) d += __shfl_xor(d, off, 16);
@@ -225,16 +233,26 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
     o1 *= alpha;
     m = mnew;
     // --- phase C: PV accumulate (thread t: dims 2*lane, position residue
-    // wid mod 4; loads independent across iterations) --------------------
-#pragma unroll 8
-    for (int k = 0; k < TILE / 4; ++k) {
-      const int po = k * 4 + wid;
-      const int p = sub0 + po;
-      if (act && p < end) {
-        const float w = stile[po];
-        const u16* vr = vbase + (size_t)p * hd + e0;
-        o0 = fmaf(w, b2f(vr[0]), o0);
-        o1 = fmaf(w, b2f(vr[1]), o1);
+    // wid mod 4).  Batches of 8 unguarded u32 loads (both dims in one
+    // load) issue together, then one wait covers all eight — the guarded
+    // 2x2-B version serialized a vmcnt(0) per position.
+#pragma unroll
+    for (int g = 0; g < TILE / 4; g += 8) {
+      unsigned int vv[8];
+      float w8[8];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const int po = (g + u) * 4 + wid;
+        const int p = sub0 + po;
+        const int pc = max(0, min(p, end - 1));
+        vv[u] = *reinterpret_cast<const unsigned int*>(
+            vbase + (size_t)pc * hd + eoff);
+        w8[u] = (act && p < end) ? stile[po] : 0.f;
+      }
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        o0 = fmaf(w8[u], b2f((u16)(vv[u] & 0xffffu)), o0);
+        o1 = fmaf(w8[u], b2f((u16)(vv[u] >> 16)), o1);
       }
     }
     __syncthreads();  // stile reused next sub-tile
@@ -281,12 +299,18 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
   float L = 0.f;
   for (int c = 0; c < nchunk; ++c)
     if (so[0][c] != -INFINITY) L += so[1][c] * __expf(so[0][c] - M);
+  // per-chunk weights once into LDS; an empty chunk has m = -INF and a
+  // ZERO partial (its o-sums were stored as 0.0), so its weight
+  // __expf(-INF - M) = 0 makes the unguarded load contribute exactly 0 —
+  // no per-chunk guard, so the nchunk sc1 loads below issue in parallel
+  // instead of one execz-guarded vmcnt(0) round trip each.
+  if (t < nchunk) so[2][t] = __expf(so[0][t] - M);
+  __syncthreads();
   for (int d = t; d < hd; d += blockDim.x) {
     float o = 0.f;
 #pragma unroll 4
     for (int c = 0; c < nchunk; ++c)
-      if (so[0][c] != -INFINITY)
-        o += WS_LOAD(&base[c * (hd + 2) + d]) * __expf(so[0][c] - M);
+      o += WS_LOAD(&base[c * (hd + 2) + d]) * so[2][c];
     outbuf[(size_t)h * hd + d] = f2b(o / L);
   }
 #undef WS_STORE
