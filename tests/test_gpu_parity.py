@@ -450,7 +450,7 @@ def test_adaptive_nchunk_recapture():
                                   size=990).astype(np.uint32)
             first = eng.prefill(prompt)
             # 140 graph-replayed steps: host_pos crosses 1024 at the s=63
-            # sync point -> re-capture with nchunk 24
+            # sync point -> re-capture with a larger chunk count
             toks = eng.decode(140)
             seq = np.concatenate([prompt, [first], toks[:-1]]).astype(
                 np.uint32)
