@@ -189,19 +189,30 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
   // zero beyond hd, so out-of-range lanes contribute exact zeros.
   const int doff = min(dgrp * 8, hd - 8);
   const int eoff = min(e0, hd - 2);
+  // K rows for the CURRENT sub-tile live in registers (ka) so the next
+  // sub-tile's loads can issue during phase C and ride the memory system
+  // behind C's loads — registers survive the barriers, and vmcnt retires
+  // in FIFO order so C's counted waits drain the prefetch for free.
+  short8 ka[TILE / 16];
+  auto issue_k = [&](int s0) {
+#pragma unroll
+    for (int pass = 0; pass < TILE / 16; ++pass) {
+      const int pc = max(0, min(s0 + pass * 16 + (t >> 4), end - 1));
+      ka[pass] =
+          *reinterpret_cast<const short8*>(kbase + (size_t)pc * hd + doff);
+    }
+  };
+  if (start < end) issue_k(start);
   for (int sub0 = start; sub0 < end; sub0 += TILE) {
     // --- phase A: scores for [sub0, sub0+TILE) --------------------------
     // 256 threads = 16 positions per pass (16 lanes per position, each
-    // loading 16 B of the K row)
+    // holding 16 B of the K row in ka)
 #pragma unroll
     for (int pass = 0; pass < TILE / 16; ++pass) {
       const int p = sub0 + pass * 16 + (t >> 4);
-      const int pc = max(0, min(p, end - 1));
-      const u16* kr = kbase + (size_t)pc * hd + doff;
-      short8 kv8 = *reinterpret_cast<const short8*>(kr);
       float d = 0.f;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) d = fmaf(b2f((u16)kv8[j]), qa[j], d);
+      for (int j = 0; j < 8; ++j) d = fmaf(b2f((u16)ka[pass][j]), qa[j], d);
       // reduce across the 16 lanes of this position
 #pragma unroll
       for (int off = 8; off >= 1; off >>= 1) d += __shfl_xor(d, off, 16);
@@ -232,27 +243,30 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
     o0 *= alpha;
     o1 *= alpha;
     m = mnew;
+    // next sub-tile's K loads issue here, ahead of phase C's own loads —
+    // C's waits retire them for free (FIFO vmcnt), so phase A of the next
+    // iteration starts with its data already on chip
+    if (sub0 + TILE < end) issue_k(sub0 + TILE);
     // --- phase C: PV accumulate (thread t: dims 2*lane, position residue
-    // wid mod 4).  Batches of 8 unguarded u32 loads (both dims in one
-    // load) issue together, then one wait covers all eight — the guarded
-    // 2x2-B version serialized a vmcnt(0) per position.
+    // wid mod 4).  Batches of 16 unguarded u32 loads (both dims in one
+    // load) issue together, then one wait covers all sixteen; the softmax
+    // weight is read from LDS at consume time (cheap, and keeps the
+    // register count inside 3-waves/SIMD occupancy).
 #pragma unroll
-    for (int g = 0; g < TILE / 4; g += 8) {
-      unsigned int vv[8];
-      float w8[8];
+    for (int g = 0; g < TILE / 4; g += 16) {
+      unsigned int vv[16];
 #pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        const int po = (g + u) * 4 + wid;
-        const int p = sub0 + po;
-        const int pc = max(0, min(p, end - 1));
+      for (int u = 0; u < 16; ++u) {
+        const int pc = max(0, min(sub0 + (g + u) * 4 + wid, end - 1));
         vv[u] = *reinterpret_cast<const unsigned int*>(
             vbase + (size_t)pc * hd + eoff);
-        w8[u] = (act && p < end) ? stile[po] : 0.f;
       }
 #pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        o0 = fmaf(w8[u], b2f((u16)(vv[u] & 0xffffu)), o0);
-        o1 = fmaf(w8[u], b2f((u16)(vv[u] >> 16)), o1);
+      for (int u = 0; u < 16; ++u) {
+        const int po = (g + u) * 4 + wid;
+        const float w = (act && sub0 + po < end) ? stile[po] : 0.f;
+        o0 = fmaf(w, b2f((u16)(vv[u] & 0xffffu)), o0);
+        o1 = fmaf(w, b2f((u16)(vv[u] >> 16)), o1);
       }
     }
     __syncthreads();  // stile reused next sub-tile
