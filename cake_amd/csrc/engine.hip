@@ -1130,15 +1130,15 @@ extern "C" int cake_hip_decode(cake_engine* e, int steps,
   if (e->host_pos + steps > e->max_seq)
     return set_err(5, "decode exceeds max_seq");
   HIP_TRY(hipMemsetAsync(e->dev_step, 0, 4, e->stream));
-  // Split-KV chunk count for decode attention, by context length (measured
-  // optima on MI355X with the unguarded-load kernel: ctx 2048 -> 16~24,
-  // 3960 -> 16, 7900/15800 -> 24; the guarded kernel wanted more chunks).
+  // Split-KV chunk count for decode attention, by context length.  With
+  // the K-prefetch kernel the curve is nearly flat (8..24 within 2%);
+  // measured optima: ctx 2048 -> 8~12, 3960 -> 12~16, 7900 -> 16~24.
   // The graph bakes the grid in, so when the context grows across a
   // threshold the graph is dropped and lazily re-captured with the new
   // chunk count (capture costs ~1 step, amortized over thousands).
   const bool nchunk_fixed = getenv("CAKE_NCHUNK") != nullptr;
   auto want_nchunk = [](int pos) {
-    return pos < 1024 ? 8 : pos < 6144 ? 16 : 24;
+    return pos < 1024 ? 8 : pos < 6144 ? 12 : 16;
   };
   bool graph_ok = e->use_graph() && e->world == 1 && !e->st.on;
   if (!nchunk_fixed) {
