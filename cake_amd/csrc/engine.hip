@@ -340,21 +340,34 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
   const ModelConfig& c = e->c;
   const int H = c.hidden, I = c.inter, hd = c.hd();
   const int Sq = c.sq(), Nq = c.nqkv();
-  {  // rms_1 fused into the qkv projection (GEMV, x kept in registers)
-    double wb = (double)Nq * H * (c.fp8 ? 1 : 2);
+  // Full-rotation models without qk-norm (llama family) take the fused
+  // rms_norm -> qkv GEMV -> rope -> KV-store kernel: one launch instead of
+  // two, with the rope applied to the dot-product sums in LDS.
+  const bool fused_qkv_rope =
+      !c.fp8 && !l.qnorm && !l.knorm && hd % 4 == 0 && (Nq % 4) == 0;
+  if (fused_qkv_rope) {
+    double wb = (double)Nq * H * 2;
     StatScope ss(e, "gemv_qkv", wb + 2.0 * H * 2 + Nq * 2, 2.0 * Nq * H);
-    if (c.fp8)
-      launch_gemv_fp8(l.wqkv8, l.sqkv, e->x, e->qkv, nullptr, l.rms1,
-                      c.rms_eps, Nq, H, 0, e->stream);
-    else
-      launch_gemv(l.wqkv, e->x, e->qkv, nullptr, l.rms1, c.rms_eps, Nq, H, 0,
-                  e->stream);
-  }
-  {  // [Qwen3 QK-norm, attention.rs:202-215, fused +] rope + KV store
-    StatScope ss(e, "rope_store", (double)Nq * hd * 0, 0);
-    launch_rope_store_decode(e->qkv, l.kc, l.vc, l.vtc, e->cos_t, e->sin_t,
-                             e->dev_pos, c.nh, c.nkv, hd, hd, e->max_seq,
-                             l.qnorm, l.knorm, c.rms_eps, e->stream);
+    launch_gemv_qkv_rope(l.wqkv, e->x, e->qkv, l.rms1, c.rms_eps, l.kc,
+                         l.vc, l.vtc, e->cos_t, e->sin_t, e->dev_pos, c.nh,
+                         c.nkv, hd, e->max_seq, H, e->stream);
+  } else {
+    {  // rms_1 fused into the qkv projection (GEMV, x kept in registers)
+      double wb = (double)Nq * H * (c.fp8 ? 1 : 2);
+      StatScope ss(e, "gemv_qkv", wb + 2.0 * H * 2 + Nq * 2, 2.0 * Nq * H);
+      if (c.fp8)
+        launch_gemv_fp8(l.wqkv8, l.sqkv, e->x, e->qkv, nullptr, l.rms1,
+                        c.rms_eps, Nq, H, 0, e->stream);
+      else
+        launch_gemv(l.wqkv, e->x, e->qkv, nullptr, l.rms1, c.rms_eps, Nq, H,
+                    0, e->stream);
+    }
+    {  // [Qwen3 QK-norm, attention.rs:202-215, fused +] rope + KV store
+      StatScope ss(e, "rope_store", (double)Nq * hd * 0, 0);
+      launch_rope_store_decode(e->qkv, l.kc, l.vc, l.vtc, e->cos_t, e->sin_t,
+                               e->dev_pos, c.nh, c.nkv, hd, hd, e->max_seq,
+                               l.qnorm, l.knorm, c.rms_eps, e->stream);
+    }
   }
   {  // decode attention over the cache (single launch, split-KV combine)
     double kvbytes = 2.0 * (e->host_pos + 1) * c.skv() * 2;

@@ -29,6 +29,11 @@ void launch_gemv_gateup(const u16* W, const u16* x, u16* out, const u16* nw,
 void launch_gemv_res_splitk(const u16* W, const u16* x, u16* out,
                             const u16* res, float* ws, u32* cnt, int N,
                             int K, hipStream_t s);
+void launch_gemv_qkv_rope(const u16* W, const u16* x, u16* out, const u16* nw,
+                          float eps, u16* kc, u16* vc, u16* vtc,
+                          const float* cost, const float* sint,
+                          const int* pos, int nh, int nkv, int hd,
+                          int max_seq, int K, hipStream_t s);
 void launch_gemv_fp8(const unsigned char* W, const float* sc, const u16* x,
                      void* out, const u16* res, const u16* nw, float eps,
                      int N, int K, int epi, hipStream_t s);

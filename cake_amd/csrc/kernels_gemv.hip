@@ -209,6 +209,218 @@ __global__ __launch_bounds__(256) void k_gemv_stream(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused [rms_norm ->] qkv GEMV -> rope -> KV-cache store, for the decode
+// step of full-rotation models without qk-norm (llama family: rd == hd).
+// Replaces the k_gemv_reg<.,0,true,.> + k_rope_store_decode launch pair —
+// one less kernel per layer, and the rope math runs on the dot-product
+// sums already sitting in LDS.
+//
+// Row mapping: a rope pair is (i, i + hd/2) within a head, so a q/k block
+// owns TWO pairs {base+i0, base+i0+1, base+i0+hd/2, base+i0+hd/2+1}
+// (i0 = 2*(b % (hd/4)), base = the head's first row).  V rows need no
+// rotation and map consecutively.  Epilogue: q rows rope'd into the qkv
+// activation buffer (attention reads them there), k rows rope'd into
+// kc[kvh][p], v rows into vc[kvh][p] + the transposed vtc image.
+// (kernels_attn.hip:31-76 is the unfused reference semantics.)
+// ---------------------------------------------------------------------------
+template <int KB>
+__global__ __launch_bounds__(256) void k_gemv_qkv_rope(
+    const u16* __restrict__ W, const u16* __restrict__ x,
+    u16* __restrict__ out, const u16* __restrict__ nw, float eps,
+    u16* __restrict__ kc, u16* __restrict__ vc, u16* __restrict__ vtc,
+    const float* __restrict__ cost, const float* __restrict__ sint,
+    const int* __restrict__ pos, int nh, int nkv, int hd, int max_seq,
+    int K) {
+  const int t = threadIdx.x;
+  const int wid = t / WAVE, lane = t % WAVE;
+  const int half = hd / 2;
+  const int hblk = hd / 4;               // pair-blocks per q/k head
+  const int nqk = (nh + nkv) * hblk;     // q/k blocks ahead of v blocks
+  const int b = blockIdx.x;
+  // rows[] = the 4 output rows this block computes
+  int rows[4];
+  bool isv;
+  int head = 0, i0 = 0;
+  if (b < nqk) {
+    isv = false;
+    head = b / hblk;
+    i0 = (b % hblk) * 2;
+    const int base = head * hd;          // q heads then k heads, contiguous
+    rows[0] = base + i0;
+    rows[1] = base + i0 + 1;
+    rows[2] = base + i0 + half;
+    rows[3] = base + i0 + half + 1;
+  } else {
+    isv = true;
+    const int vrow0 = (nh + nkv) * hd + (b - nqk) * 4;
+    rows[0] = vrow0;
+    rows[1] = vrow0 + 1;
+    rows[2] = vrow0 + 2;
+    rows[3] = vrow0 + 3;
+  }
+  __shared__ float red[4][4];
+
+  // same load-issue order as k_gemv_reg: x, norm weights, then row weights
+  short8 xpre[KB];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    if (k0 < K) xpre[i] = *reinterpret_cast<const short8*>(x + k0);
+  }
+  short8 nwpre[KB];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    if (k0 < K) nwpre[i] = *reinterpret_cast<const short8*>(nw + k0);
+  }
+  short8 wpre[KB <= 2 ? 4 : 1][KB <= 2 ? KB : 1];
+  if (KB <= 2) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+#pragma unroll
+      for (int i = 0; i < KB; ++i) {
+        const int k0 = i * 2048 + t * 8;
+        if (k0 < K)
+          wpre[r][i] = ntload8(W + (size_t)rows[r] * K + k0);
+      }
+  }
+
+  float xr[KB * 8];
+#pragma unroll
+  for (int i = 0; i < KB; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    if (k0 < K) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xpre[i][j]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = 0.f;
+    }
+  }
+  {  // fused rms_norm (bit-exact with the unfused pair, see k_gemv_reg)
+    float ss = 0.f;
+#pragma unroll
+    for (int i = 0; i < KB * 8; ++i) ss += xr[i] * xr[i];
+    ss = wave_sum(ss);
+    if (lane == 0) red[0][wid] = ss;
+    __syncthreads();
+    const float scale =
+        rsqrtf((red[0][0] + red[0][1] + red[0][2] + red[0][3]) / (float)K +
+               eps);
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          xr[i * 8 + j] =
+              b2f(f2b(xr[i * 8 + j] * scale * b2f((u16)nwpre[i][j])));
+      }
+    }
+  }
+
+  float acc[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) acc[r] = 0.f;
+  if (KB <= 2) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+#pragma unroll
+      for (int i = 0; i < KB; ++i) {
+        const int k0 = i * 2048 + t * 8;
+        if (k0 < K) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            acc[r] = fmaf(b2f((u16)wpre[r][i][j]), xr[i * 8 + j], acc[r]);
+        }
+      }
+  } else {
+#pragma unroll
+    for (int i = 0; i < KB; ++i) {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) {
+        short8 wv[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          wv[r] = ntload8(W + (size_t)rows[r] * K + k0);
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            acc[r] = fmaf(b2f((u16)wv[r][j]), xr[i * 8 + j], acc[r]);
+        }
+      }
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    float v = wave_sum(acc[r]);
+    if (lane == 0) red[r][wid] = v;
+  }
+  __syncthreads();
+  if (t == 0) {
+    float s[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      s[r] = red[r][0] + red[r][1] + red[r][2] + red[r][3];
+    const int p = *pos;
+    if (isv) {
+      const int kvh = (rows[0] - (nh + nkv) * hd) / hd;
+      const int d0 = rows[0] % hd;
+      u16* dst = vc + ((size_t)kvh * max_seq + p) * hd + d0;
+      u16* dstt = vtc + (size_t)kvh * hd * max_seq + (size_t)d0 * max_seq + p;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const u16 bv = f2b(s[r]);
+        dst[r] = bv;
+        dstt[(size_t)r * max_seq] = bv;
+      }
+    } else {
+      const float* c = cost + (size_t)p * half + i0;
+      const float* sn = sint + (size_t)p * half + i0;
+      u16 o[4];
+      o[0] = f2b(s[0] * c[0] - s[2] * sn[0]);
+      o[1] = f2b(s[1] * c[1] - s[3] * sn[1]);
+      o[2] = f2b(s[2] * c[0] + s[0] * sn[0]);
+      o[3] = f2b(s[3] * c[1] + s[1] * sn[1]);
+      if (head < nh) {  // q -> activation buffer, in place
+        u16* q = out + (size_t)head * hd + i0;
+        q[0] = o[0];
+        q[1] = o[1];
+        q[half] = o[2];
+        q[half + 1] = o[3];
+      } else {          // k -> cache slot p
+        const int kvh = head - nh;
+        u16* dst = kc + ((size_t)kvh * max_seq + p) * hd + i0;
+        dst[0] = o[0];
+        dst[1] = o[1];
+        dst[half] = o[2];
+        dst[half + 1] = o[3];
+      }
+    }
+  }
+}
+
+void launch_gemv_qkv_rope(const u16* W, const u16* x, u16* out, const u16* nw,
+                          float eps, u16* kc, u16* vc, u16* vtc,
+                          const float* cost, const float* sint,
+                          const int* pos, int nh, int nkv, int hd,
+                          int max_seq, int K, hipStream_t s) {
+  const int nblk = (nh + 2 * nkv) * hd / 4;
+  dim3 grid(nblk);
+#define QKR(KB)                                                             \
+  hipLaunchKernelGGL((k_gemv_qkv_rope<KB>), grid, dim3(256), 0, s, W, x,    \
+                     out, nw, eps, kc, vc, vtc, cost, sint, pos, nh, nkv,   \
+                     hd, max_seq, K)
+  if (K <= 2048) QKR(1);
+  else if (K <= 4096) QKR(2);
+  else if (K <= 8192) QKR(4);
+  else QKR(8);
+#undef QKR
+}
+
 // fused [rms_norm ->] gate_up GEMV -> silu_mul: block computes
 // out[i] = silu(g_i) * u_i for 8 channels, g_i = dot(W[i,:], xn),
 // u_i = dot(W[i+I,:], xn)  (mlp.rs:21-31 + fused gate_up of mlp.rs:38-46)
