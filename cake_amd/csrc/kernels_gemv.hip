@@ -31,18 +31,37 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
   // weights are consumed after the reduction barrier; the row weights
   // last, in the FMA phase.  All three groups are in flight together, so
   // a drain before sumsq costs one memory latency, not three.
+  // All loads below are UNGUARDED with clamped addresses (same fix as the
+  // attention kernel): an `if (row < N && k0 < K)` around a load costs
+  // hipcc an execz block + vmcnt(0) drain per unrolled iteration.  Out-of-
+  // range k contributes 0 through xr (zeroed beyond K); out-of-range rows
+  // compute garbage sums that the guarded epilogue never stores.
+  // KB == 1 can be ragged (K < 2048, e.g. qwen3-0.6b H=1024): there the
+  // guard SKIPS loads for fully out-of-range threads, which beats the
+  // clamped duplicate loads (-2.4% A/B'd on qwen3-0.6b); KB >= 2 shapes
+  // win with clamping (no execz blocks, counted waits).
   short8 xpre[KB];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
-    const int k0 = i * 2048 + t * 8;
-    if (k0 < K) xpre[i] = *reinterpret_cast<const short8*>(x + k0);
+    if constexpr (KB >= 2) {
+      const int k0c = max(0, min(i * 2048 + t * 8, K - 8));
+      xpre[i] = *reinterpret_cast<const short8*>(x + k0c);
+    } else {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) xpre[i] = *reinterpret_cast<const short8*>(x + k0);
+    }
   }
   short8 nwpre[NORM ? KB : 1];
   if (NORM) {
 #pragma unroll
     for (int i = 0; i < KB; ++i) {
-      const int k0 = i * 2048 + t * 8;
-      if (k0 < K) nwpre[i] = *reinterpret_cast<const short8*>(nw + k0);
+      if constexpr (KB >= 2) {
+        const int k0c = max(0, min(i * 2048 + t * 8, K - 8));
+        nwpre[i] = *reinterpret_cast<const short8*>(nw + k0c);
+      } else {
+        const int k0 = i * 2048 + t * 8;
+        if (k0 < K) nwpre[i] = *reinterpret_cast<const short8*>(nw + k0);
+      }
     }
   }
   // For short K, issue ALL weight loads too — they stay in flight across
@@ -54,24 +73,27 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
     for (int r = 0; r < ROWS; ++r)
 #pragma unroll
       for (int i = 0; i < KB; ++i) {
-        const int k0 = i * 2048 + t * 8;
-        if (row0 + r < N && k0 < K)
-          wpre[r][i] = ntload8(W + (size_t)(row0 + r) * K + k0);
+        if constexpr (KB >= 2) {
+          const int k0c = max(0, min(i * 2048 + t * 8, K - 8));
+          const int rc = min(row0 + r, N - 1);
+          wpre[r][i] = ntload8(W + (size_t)rc * K + k0c);
+        } else {
+          const int k0 = i * 2048 + t * 8;
+          if (row0 + r < N && k0 < K)
+            wpre[r][i] = ntload8(W + (size_t)(row0 + r) * K + k0);
+        }
       }
   }
 
-  // phase 1: x -> f32, optionally fused rms_norm
+  // phase 1: x -> f32, optionally fused rms_norm (k >= K slots zeroed so
+  // the clamped duplicate loads contribute nothing anywhere downstream)
   float xr[KB * 8];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
-    const int k0 = i * 2048 + t * 8;
-    if (k0 < K) {
+    const bool in = i * 2048 + t * 8 < K;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xpre[i][j]);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = 0.f;
-    }
+    for (int j = 0; j < 8; ++j)
+      xr[i * 8 + j] = in ? b2f((u16)xpre[i][j]) : 0.f;
   }
   if (NORM) {
     float ss = 0.f;
@@ -107,8 +129,7 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
     for (int r = 0; r < ROWS; ++r)
 #pragma unroll
       for (int i = 0; i < KB; ++i) {
-        const int k0 = i * 2048 + t * 8;
-        if (row0 + r < N && k0 < K) {
+        if (KB >= 2 || (row0 + r < N && i * 2048 + t * 8 < K)) {
 #pragma unroll
           for (int j = 0; j < 8; ++j)
             acc[r] = fmaf(b2f((u16)wpre[r][i][j]), xr[i * 8 + j], acc[r]);
@@ -117,19 +138,16 @@ __global__ __launch_bounds__(256) void k_gemv_reg(
   } else {
 #pragma unroll
     for (int i = 0; i < KB; ++i) {
-      const int k0 = i * 2048 + t * 8;
-      if (k0 < K) {
-        short8 wv[ROWS];
+      const int k0c = max(0, min(i * 2048 + t * 8, K - 8));
+      short8 wv[ROWS];
 #pragma unroll
-        for (int r = 0; r < ROWS; ++r)
-          if (row0 + r < N) wv[r] = ntload8(W + (size_t)(row0 + r) * K + k0);
+      for (int r = 0; r < ROWS; ++r)
+        wv[r] = ntload8(W + (size_t)min(row0 + r, N - 1) * K + k0c);
 #pragma unroll
-        for (int r = 0; r < ROWS; ++r)
-          if (row0 + r < N) {
+      for (int r = 0; r < ROWS; ++r) {
 #pragma unroll
-            for (int j = 0; j < 8; ++j)
-              acc[r] = fmaf(b2f((u16)wv[r][j]), xr[i * 8 + j], acc[r]);
-          }
+        for (int j = 0; j < 8; ++j)
+          acc[r] = fmaf(b2f((u16)wv[r][j]), xr[i * 8 + j], acc[r]);
       }
     }
   }
@@ -265,14 +283,14 @@ __global__ __launch_bounds__(256) void k_gemv_qkv_rope(
   short8 xpre[KB];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
-    const int k0 = i * 2048 + t * 8;
-    if (k0 < K) xpre[i] = *reinterpret_cast<const short8*>(x + k0);
+    const int k0c = max(0, min(i * 2048 + t * 8, K - 8));
+    xpre[i] = *reinterpret_cast<const short8*>(x + k0c);
   }
   short8 nwpre[KB];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
-    const int k0 = i * 2048 + t * 8;
-    if (k0 < K) nwpre[i] = *reinterpret_cast<const short8*>(nw + k0);
+    const int k0c = max(0, min(i * 2048 + t * 8, K - 8));
+    nwpre[i] = *reinterpret_cast<const short8*>(nw + k0c);
   }
   short8 wpre[KB <= 2 ? 4 : 1][KB <= 2 ? KB : 1];
   if (KB <= 2) {
@@ -280,23 +298,18 @@ __global__ __launch_bounds__(256) void k_gemv_qkv_rope(
     for (int r = 0; r < 4; ++r)
 #pragma unroll
       for (int i = 0; i < KB; ++i) {
-        const int k0 = i * 2048 + t * 8;
-        if (k0 < K)
-          wpre[r][i] = ntload8(W + (size_t)rows[r] * K + k0);
+        const int k0c = max(0, min(i * 2048 + t * 8, K - 8));
+        wpre[r][i] = ntload8(W + (size_t)rows[r] * K + k0c);
       }
   }
 
   float xr[KB * 8];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
-    const int k0 = i * 2048 + t * 8;
-    if (k0 < K) {
+    const bool in = i * 2048 + t * 8 < K;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xpre[i][j]);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = 0.f;
-    }
+    for (int j = 0; j < 8; ++j)
+      xr[i * 8 + j] = in ? b2f((u16)xpre[i][j]) : 0.f;
   }
   {  // fused rms_norm (bit-exact with the unfused pair, see k_gemv_reg)
     float ss = 0.f;
@@ -329,28 +342,23 @@ __global__ __launch_bounds__(256) void k_gemv_qkv_rope(
     for (int r = 0; r < 4; ++r)
 #pragma unroll
       for (int i = 0; i < KB; ++i) {
-        const int k0 = i * 2048 + t * 8;
-        if (k0 < K) {
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            acc[r] = fmaf(b2f((u16)wpre[r][i][j]), xr[i * 8 + j], acc[r]);
-        }
+        for (int j = 0; j < 8; ++j)
+          acc[r] = fmaf(b2f((u16)wpre[r][i][j]), xr[i * 8 + j], acc[r]);
       }
   } else {
 #pragma unroll
     for (int i = 0; i < KB; ++i) {
-      const int k0 = i * 2048 + t * 8;
-      if (k0 < K) {
-        short8 wv[4];
+      const int k0c = max(0, min(i * 2048 + t * 8, K - 8));
+      short8 wv[4];
 #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          wv[r] = ntload8(W + (size_t)rows[r] * K + k0);
+      for (int r = 0; r < 4; ++r)
+        wv[r] = ntload8(W + (size_t)rows[r] * K + k0c);
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
+      for (int r = 0; r < 4; ++r) {
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            acc[r] = fmaf(b2f((u16)wv[r][j]), xr[i * 8 + j], acc[r]);
-        }
+        for (int j = 0; j < 8; ++j)
+          acc[r] = fmaf(b2f((u16)wv[r][j]), xr[i * 8 + j], acc[r]);
       }
     }
   }
@@ -436,31 +444,39 @@ __global__ __launch_bounds__(256) void k_gemv_gateup(
 
   // load issue order = consumer order (see k_gemv_reg): x first (feeds
   // sumsq), then norm weights (consumed after the barrier)
+  // unguarded clamped loads for KB >= 2, guarded for ragged KB == 1
+  // (see k_gemv_reg)
   short8 xpre[KB];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
-    const int k0 = i * 2048 + t * 8;
-    if (k0 < K) xpre[i] = *reinterpret_cast<const short8*>(x + k0);
+    if constexpr (KB >= 2) {
+      const int k0c = max(0, min(i * 2048 + t * 8, K - 8));
+      xpre[i] = *reinterpret_cast<const short8*>(x + k0c);
+    } else {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) xpre[i] = *reinterpret_cast<const short8*>(x + k0);
+    }
   }
   short8 nwpre[NORM ? KB : 1];
   if (NORM) {
 #pragma unroll
     for (int i = 0; i < KB; ++i) {
-      const int k0 = i * 2048 + t * 8;
-      if (k0 < K) nwpre[i] = *reinterpret_cast<const short8*>(nw + k0);
+      if constexpr (KB >= 2) {
+        const int k0c = max(0, min(i * 2048 + t * 8, K - 8));
+        nwpre[i] = *reinterpret_cast<const short8*>(nw + k0c);
+      } else {
+        const int k0 = i * 2048 + t * 8;
+        if (k0 < K) nwpre[i] = *reinterpret_cast<const short8*>(nw + k0);
+      }
     }
   }
   float xr[KB * 8];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
-    const int k0 = i * 2048 + t * 8;
-    if (k0 < K) {
+    const bool in = i * 2048 + t * 8 < K;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = b2f((u16)xpre[i][j]);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) xr[i * 8 + j] = 0.f;
-    }
+    for (int j = 0; j < 8; ++j)
+      xr[i * 8 + j] = in ? b2f((u16)xpre[i][j]) : 0.f;
   }
   if (NORM) {
     float ss = 0.f;
@@ -490,24 +506,43 @@ __global__ __launch_bounds__(256) void k_gemv_gateup(
   for (int r = 0; r < ROWS; ++r) accg[r] = accu[r] = 0.f;
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
-    const int k0 = i * 2048 + t * 8;
-    if (k0 < K) {
+    if constexpr (KB >= 2) {
+      const int k0c = max(0, min(i * 2048 + t * 8, K - 8));
       short8 gv[ROWS], uv[ROWS];
 #pragma unroll
-      for (int r = 0; r < ROWS; ++r)
-        if (c0 + r < I) {
-          gv[r] = ntload8(W + (size_t)(c0 + r) * K + k0);
-          uv[r] = ntload8(W + (size_t)(c0 + r + I) * K + k0);
+      for (int r = 0; r < ROWS; ++r) {
+        const int rc = min(c0 + r, I - 1);
+        gv[r] = ntload8(W + (size_t)rc * K + k0c);
+        uv[r] = ntload8(W + (size_t)(rc + I) * K + k0c);
+      }
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          accg[r] = fmaf(b2f((u16)gv[r][j]), xr[i * 8 + j], accg[r]);
+          accu[r] = fmaf(b2f((u16)uv[r][j]), xr[i * 8 + j], accu[r]);
         }
+      }
+    } else {
+      const int k0 = i * 2048 + t * 8;
+      if (k0 < K) {
+        short8 gv[ROWS], uv[ROWS];
 #pragma unroll
-      for (int r = 0; r < ROWS; ++r)
-        if (c0 + r < I) {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            accg[r] = fmaf(b2f((u16)gv[r][j]), xr[i * 8 + j], accg[r]);
-            accu[r] = fmaf(b2f((u16)uv[r][j]), xr[i * 8 + j], accu[r]);
+        for (int r = 0; r < ROWS; ++r)
+          if (c0 + r < I) {
+            gv[r] = ntload8(W + (size_t)(c0 + r) * K + k0);
+            uv[r] = ntload8(W + (size_t)(c0 + r + I) * K + k0);
           }
-        }
+#pragma unroll
+        for (int r = 0; r < ROWS; ++r)
+          if (c0 + r < I) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              accg[r] = fmaf(b2f((u16)gv[r][j]), xr[i * 8 + j], accg[r]);
+              accu[r] = fmaf(b2f((u16)uv[r][j]), xr[i * 8 + j], accu[r]);
+            }
+          }
+      }
     }
   }
 #pragma unroll
