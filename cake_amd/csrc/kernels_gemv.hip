@@ -185,25 +185,43 @@ __global__ __launch_bounds__(256) void k_gemv_stream(
 #pragma unroll
   for (int r = 0; r < ROWS; ++r) acc[r] = 0.f;
   const int kiter = (K + 2047) / 2048;
+  // k-outer with all rows' loads issued per iteration (ROWS+1 loads in
+  // flight instead of a serial per-row chain), unguarded with clamped
+  // addresses: out-of-range k zeroes x, out-of-range rows compute garbage
+  // the guarded epilogue never stores.  K % 8 == 0 is the call contract
+  // for this path (checked by the launcher's K % 8 dispatch).
 #pragma unroll 1
-  for (int r = 0; r < ROWS; ++r) {
-    const int row = row0 + r;
-    if (row >= N) break;
-    const u16* wr = W + (size_t)row * K;
-    float a = 0.f;
-    for (int i = 0; i < kiter; ++i) {
-      int k0 = i * 2048 + t * 8;
-      if (k0 + 8 <= K) {
-        short8 wv = ntload8(wr + k0);
-        short8 xv = *reinterpret_cast<const short8*>(x + k0);
+  for (int i = 0; i < kiter; ++i) {
+    const int k0 = i * 2048 + t * 8;
+    const int k0c = max(0, min(k0, K - 8));
+    short8 xv = *reinterpret_cast<const short8*>(x + k0c);
+    float xf[8];
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          a = fmaf(b2f((u16)wv[j]), b2f((u16)xv[j]), a);
-      } else {
-        for (int k = k0; k < K; ++k) a = fmaf(b2f(wr[k]), b2f(x[k]), a);
-      }
+    for (int j = 0; j < 8; ++j)
+      xf[j] = k0 + 8 <= K ? b2f((u16)xv[j]) : 0.f;
+    short8 wv[ROWS];
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+      wv[r] = ntload8(W + (size_t)min(row0 + r, N - 1) * K + k0c);
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        acc[r] = fmaf(b2f((u16)wv[r][j]), xf[j], acc[r]);
     }
-    acc[r] = a;
+  }
+  // scalar tail for K % 8 != 0 (op-level shapes; model dims are % 8):
+  // thread t < K % 8 handles element (K - K%8 + t) for every row; the
+  // wave/block reduction below folds it in like any other partial
+  if (K % 8) {
+    const int k = K - K % 8 + t;
+    if (t < K % 8) {
+      const float xt = b2f(x[k]);
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r)
+        if (row0 + r < N)
+          acc[r] = fmaf(b2f(W[(size_t)(row0 + r) * K + k]), xt, acc[r]);
+    }
   }
   __shared__ float red[ROWS > 4 ? ROWS : 4][4];
 #pragma unroll
@@ -576,15 +594,23 @@ __global__ __launch_bounds__(256) void k_gemv_fp8(
   const int wid = t / WAVE, lane = t % WAVE;
   __shared__ float red[ROWS > 4 ? ROWS : 4][4];
 
-  // prefetch all weight tiles (stay in flight across the x/norm phase)
+  // prefetch all weight tiles (stay in flight across the x/norm phase);
+  // unguarded clamped addresses for KB >= 2 (see k_gemv_reg: guarded
+  // loads cost an execz block + vmcnt(0) drain per unrolled iteration)
   uint4v wpre[ROWS][KB];
 #pragma unroll
   for (int r = 0; r < ROWS; ++r)
 #pragma unroll
     for (int i = 0; i < KB; ++i) {
-      const int k0 = i * 4096 + t * 16;
-      if (row0 + r < N && k0 < K)
-        wpre[r][i] = ntload16b(W + (size_t)(row0 + r) * K + k0);
+      if constexpr (KB >= 2) {
+        const int k0c = max(0, min(i * 4096 + t * 16, K - 16));
+        wpre[r][i] =
+            ntload16b(W + (size_t)min(row0 + r, N - 1) * K + k0c);
+      } else {
+        const int k0 = i * 4096 + t * 16;
+        if (row0 + r < N && k0 < K)
+          wpre[r][i] = ntload16b(W + (size_t)(row0 + r) * K + k0);
+      }
     }
 
   // x -> registers (f32), optionally fused rms_norm.  (The bf16 kernels'
@@ -597,7 +623,13 @@ __global__ __launch_bounds__(256) void k_gemv_fp8(
     const int k0 = i * 4096 + t * 16;
 #pragma unroll
     for (int half = 0; half < 2; ++half) {
-      if (k0 < K) {
+      if (KB >= 2) {
+        const int k0c = max(0, min(k0, K - 16)) + half * 8;
+        short8 xv = *reinterpret_cast<const short8*>(x + k0c);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          xr[i * 16 + half * 8 + j] = k0 < K ? b2f((u16)xv[j]) : 0.f;
+      } else if (k0 < K) {
         short8 xv = *reinterpret_cast<const short8*>(x + k0 + half * 8);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
@@ -624,7 +656,15 @@ __global__ __launch_bounds__(256) void k_gemv_fp8(
       const int k0 = i * 4096 + t * 16;
 #pragma unroll
       for (int half = 0; half < 2; ++half) {
-        if (k0 < K) {
+        if (KB >= 2) {
+          const int k0c = max(0, min(k0, K - 16)) + half * 8;
+          short8 wv = *reinterpret_cast<const short8*>(nw + k0c);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int ix = i * 16 + half * 8 + j;
+            xr[ix] = b2f(f2b(xr[ix] * scale * b2f((u16)wv[j])));
+          }
+        } else if (k0 < K) {
           short8 wv =
               *reinterpret_cast<const short8*>(nw + k0 + half * 8);
 #pragma unroll
@@ -645,10 +685,12 @@ __global__ __launch_bounds__(256) void k_gemv_fp8(
 #pragma unroll
     for (int i = 0; i < KB; ++i) {
       const int k0 = i * 4096 + t * 16;
-      if (row0 + r < N && k0 < K) {
-        // one scale_inv block covers this thread's 16 k's (k0 % 16 == 0)
-        const float s =
-            sc[(size_t)((row0 + r) / 128) * nkb + (k0 / 128)];
+      if (KB >= 2 || (row0 + r < N && k0 < K)) {
+        // one scale_inv block covers this thread's 16 k's (k0 % 16 == 0);
+        // clamped indices keep the scale read in-bounds for the dup loads
+        const int rc = min(row0 + r, N - 1);
+        const int k0c = max(0, min(k0, K - 16));
+        const float s = sc[(size_t)(rc / 128) * nkb + (k0c / 128)];
         float wd[16];
         f8x16_decode(wpre[r][i], wd);
         float a = 0.f;
@@ -692,33 +734,38 @@ __global__ __launch_bounds__(256) void k_gemv_fp8_stream(
 #pragma unroll
   for (int r = 0; r < ROWS; ++r) acc[r] = 0.f;
   const int kiter = (K + 4095) / 4096;
+  // unguarded clamped body (see k_gemv_reg): all ROWS+2 loads of an
+  // iteration issue together; out-of-range k zeroes x, out-of-range rows
+  // compute garbage the guarded epilogue never stores.  K % 16 == 0 on
+  // this path (fp8 blocks are 128-wide).
 #pragma unroll 1
   for (int i = 0; i < kiter; ++i) {
     const int k0 = i * 4096 + t * 16;
-    if (k0 >= K) continue;
+    const int k0c = max(0, min(k0, K - 16));
     float xv[16];
 #pragma unroll
     for (int half = 0; half < 2; ++half) {
-      short8 xs = *reinterpret_cast<const short8*>(x + k0 + half * 8);
+      short8 xs = *reinterpret_cast<const short8*>(x + k0c + half * 8);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) xv[half * 8 + j] = b2f((u16)xs[j]);
+      for (int j = 0; j < 8; ++j)
+        xv[half * 8 + j] = k0 < K ? b2f((u16)xs[j]) : 0.f;
     }
     uint4v wv[ROWS];
 #pragma unroll
     for (int r = 0; r < ROWS; ++r)
-      if (row0 + r < N) wv[r] = ntload16b(W + (size_t)(row0 + r) * K + k0);
+      wv[r] = ntload16b(W + (size_t)min(row0 + r, N - 1) * K + k0c);
 #pragma unroll
-    for (int r = 0; r < ROWS; ++r)
-      if (row0 + r < N) {
-        const float s = sc[(size_t)((row0 + r) / 128) * nkb + (k0 / 128)];
-        float wd[16];
-        f8x16_decode(wv[r], wd);
-        float a = 0.f;
+    for (int r = 0; r < ROWS; ++r) {
+      const float s =
+          sc[(size_t)(min(row0 + r, N - 1) / 128) * nkb + (k0c / 128)];
+      float wd[16];
+      f8x16_decode(wv[r], wd);
+      float a = 0.f;
 #pragma unroll
-        for (int j = 0; j < 16; ++j)
-          a = fmaf(wd[j], xv[j], a);
-        acc[r] = fmaf(a, s, acc[r]);
-      }
+      for (int j = 0; j < 16; ++j)
+        a = fmaf(wd[j], xv[j], a);
+      acc[r] = fmaf(a, s, acc[r]);
+    }
   }
   __shared__ float red[ROWS > 4 ? ROWS : 4][4];
 #pragma unroll
@@ -753,15 +800,20 @@ __global__ __launch_bounds__(256) void k_gemv_gateup_fp8(
   const int wid = t / WAVE, lane = t % WAVE;
   __shared__ float redg[8][4], redu[8][4];
 
-  // x -> registers; early nw/x prefetch is a measured NEGATIVE on the fp8
-  // path (occupancy, see k_gemv_fp8 note)
+  // x -> registers; unguarded clamped loads for KB >= 2 (see k_gemv_reg)
   float xr[KB * 16];
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
     const int k0 = i * 4096 + t * 16;
 #pragma unroll
     for (int half = 0; half < 2; ++half) {
-      if (k0 < K) {
+      if (KB >= 2) {
+        const int k0c = max(0, min(k0, K - 16)) + half * 8;
+        short8 xv = *reinterpret_cast<const short8*>(x + k0c);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          xr[i * 16 + half * 8 + j] = k0 < K ? b2f((u16)xv[j]) : 0.f;
+      } else if (k0 < K) {
         short8 xv = *reinterpret_cast<const short8*>(x + k0 + half * 8);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
@@ -788,7 +840,15 @@ __global__ __launch_bounds__(256) void k_gemv_gateup_fp8(
       const int k0 = i * 4096 + t * 16;
 #pragma unroll
       for (int half = 0; half < 2; ++half) {
-        if (k0 < K) {
+        if (KB >= 2) {
+          const int k0c = max(0, min(k0, K - 16)) + half * 8;
+          short8 wv = *reinterpret_cast<const short8*>(nw + k0c);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int ix = i * 16 + half * 8 + j;
+            xr[ix] = b2f(f2b(xr[ix] * scale * b2f((u16)wv[j])));
+          }
+        } else if (k0 < K) {
           short8 wv =
               *reinterpret_cast<const short8*>(nw + k0 + half * 8);
 #pragma unroll
@@ -807,20 +867,20 @@ __global__ __launch_bounds__(256) void k_gemv_gateup_fp8(
 #pragma unroll
   for (int i = 0; i < KB; ++i) {
     const int k0 = i * 4096 + t * 16;
-    if (k0 >= K) continue;
-    uint4v gv[ROWS], uv[ROWS];
+    if (KB >= 2) {
+      const int k0c = max(0, min(k0, K - 16));
+      uint4v gv[ROWS], uv[ROWS];
 #pragma unroll
-    for (int r = 0; r < ROWS; ++r)
-      if (c0 + r < I) {
-        gv[r] = ntload16b(W + (size_t)(c0 + r) * K + k0);
-        uv[r] = ntload16b(W + (size_t)(c0 + r + I) * K + k0);
+      for (int r = 0; r < ROWS; ++r) {
+        const int rc = min(c0 + r, I - 1);
+        gv[r] = ntload16b(W + (size_t)rc * K + k0c);
+        uv[r] = ntload16b(W + (size_t)(rc + I) * K + k0c);
       }
 #pragma unroll
-    for (int r = 0; r < ROWS; ++r)
-      if (c0 + r < I) {
-        const float sg = sc[(size_t)((c0 + r) / 128) * nkb + (k0 / 128)];
-        const float su =
-            sc[(size_t)((c0 + r + I) / 128) * nkb + (k0 / 128)];
+      for (int r = 0; r < ROWS; ++r) {
+        const int rc = min(c0 + r, I - 1);
+        const float sg = sc[(size_t)(rc / 128) * nkb + (k0c / 128)];
+        const float su = sc[(size_t)((rc + I) / 128) * nkb + (k0c / 128)];
         float gd[16], ud[16];
         f8x16_decode(gv[r], gd);
         f8x16_decode(uv[r], ud);
@@ -833,6 +893,33 @@ __global__ __launch_bounds__(256) void k_gemv_gateup_fp8(
         accg[r] = fmaf(g, sg, accg[r]);
         accu[r] = fmaf(u, su, accu[r]);
       }
+    } else if (k0 < K) {
+      uint4v gv[ROWS], uv[ROWS];
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r)
+        if (c0 + r < I) {
+          gv[r] = ntload16b(W + (size_t)(c0 + r) * K + k0);
+          uv[r] = ntload16b(W + (size_t)(c0 + r + I) * K + k0);
+        }
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r)
+        if (c0 + r < I) {
+          const float sg = sc[(size_t)((c0 + r) / 128) * nkb + (k0 / 128)];
+          const float su =
+              sc[(size_t)((c0 + r + I) / 128) * nkb + (k0 / 128)];
+          float gd[16], ud[16];
+          f8x16_decode(gv[r], gd);
+          f8x16_decode(uv[r], ud);
+          float g = 0.f, u = 0.f;
+#pragma unroll
+          for (int j = 0; j < 16; ++j) {
+            g = fmaf(gd[j], xr[i * 16 + j], g);
+            u = fmaf(ud[j], xr[i * 16 + j], u);
+          }
+          accg[r] = fmaf(g, sg, accg[r]);
+          accu[r] = fmaf(u, su, accu[r]);
+        }
+    }
   }
 #pragma unroll
   for (int r = 0; r < ROWS; ++r) {
