@@ -367,12 +367,16 @@ __global__ __launch_bounds__(512) void k_gemm_256(
     const u16* bp = piece(1, buf, wn >> 1);          // this wave's B half
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      if (tau + 1 < NT) {
-        stage_op(ks, tau + 1);                       // ks0: A, ks1: B
-        // tile tau fully resident once all but the 4 just-issued retire
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-      } else {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // last tile
+      if (tau + 1 < NT) stage_op(ks, tau + 1);       // ks0: A, ks1: B
+      if (ks == 0) {
+        // tile tau fully resident once all but the 4 just-issued retire;
+        // ks1 needs NO wait (tau drained at ks0, and waiting would force
+        // the just-staged tau+1 pieces to land — serializing the
+        // prefetch instead of letting it ride through ks1's compute)
+        if (tau + 1 < NT)
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
       __builtin_amdgcn_s_barrier();
       bf16x8 af[8], bf[4];
@@ -403,6 +407,144 @@ __global__ __launch_bounds__(512) void k_gemm_256(
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // quiesce before exit
 
   // epilogue (C/D map: col = lane&15, row = (lane>>4)*4 + reg)
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wm * 128 + mi * 16 + (lane / 16) * 4 + r;
+        const int col = n0 + wn * 64 + ni * 16 + (lane % 16);
+        if (row < M && col < N) {
+          float v = acc[mi][ni][r];
+          if (EPI == 1) v += b2f(res[(size_t)row * N + col]);
+          C[(size_t)row * N + col] = f2b(v);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// 256^2 fine-phase variant (guide §5.5 8-phase template direction): the
+// K-tile's 64 MFMA are split into FOUR 16-MFMA phases (mi-half x K-half
+// quadrants), each phase staging one (op, row-half) piece of the next tile
+// (2 glds) so the global traffic interleaves 1:1 with compute, and the
+// per-tile wait is a single counted vmcnt(2) (the two loads just issued)
+// instead of two vmcnt(4)s.  B fragments are register-reused across the
+// two mi-half phases of a K-half (af[4]+bf[4] live = 32 VGPRs vs 48).
+// "The per-phase interleave is the lever" — coarse phase splits measured
+// -7..-27% in the guide's ablation.
+// ---------------------------------------------------------------------------
+template <int EPI>
+__global__ __launch_bounds__(512) void k_gemm_256p(
+    const u16* __restrict__ A, const u16* __restrict__ W, u16* __restrict__ C,
+    const u16* __restrict__ res, int M, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  auto piece = [&](int op, int buf, int half) -> u16* {
+    return reinterpret_cast<u16*>(smem + op * 65536 +
+                                  (buf * 2 + half) * 16384);
+  };
+  int mt, nt;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    const int orig = blockIdx.x + gridDim.x * blockIdx.y;
+    const int q = nwg / 8, rr = nwg % 8;
+    const int xcd = orig % 8, idx = orig / 8;
+    const int wgid =
+        (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+    mt = wgid % gridDim.x;
+    nt = wgid / gridDim.x;
+  }
+  const int m0 = mt * 256, n0 = nt * 256;
+  const int t = threadIdx.x;
+  const int w = t / WAVE, lane = t % WAVE;
+  const int wm = w >> 2, wn = w & 3;
+
+  // stage ONE (op, half) piece of tile tau: 2 glds per wave
+  auto stage_piece = [&](int op, int half, int tau) {
+    const u16* G = op == 0 ? A : W;
+    const int rows = op == 0 ? M : N;
+    const int base = op == 0 ? m0 : n0;
+    const int buf = tau & 1;
+    const int kt = tau * 64;
+    u16* dst = piece(op, buf, half);
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int rloc = (w * 2 + j) * 8 + lane / 8;  // 0..127
+      const int grow = min(base + half * 128 + rloc, rows - 1);
+      const int u = (lane % 8) ^ (rloc & 7);
+      const u16* src = G + (size_t)grow * K + kt + u * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(uintptr_t)src,
+          (__attribute__((address_space(3))) void*)(uintptr_t)(
+              dst + (size_t)(w * 2 + j) * 8 * 64),
+          16, 0, 0);
+    }
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int NT = K / 64;
+  // prologue: all four pieces of tile 0
+  stage_piece(0, 0, 0);
+  stage_piece(0, 1, 0);
+  stage_piece(1, 0, 0);
+  stage_piece(1, 1, 0);
+  for (int tau = 0; tau < NT; ++tau) {
+    const int buf = tau & 1;
+    const u16* ap = piece(0, buf, wm);
+    const u16* bp = piece(1, buf, wn >> 1);
+    bf16x8 bfr[4];
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+#pragma unroll
+      for (int q = 0; q < 2; ++q) {
+        const int p = ks * 2 + q;
+        if (tau + 1 < NT) stage_piece(p >> 1, p & 1, tau + 1);
+        if (p == 0) {
+          // tile tau resident once everything but the piece just issued
+          // has retired (per-wave FIFO vmcnt)
+          if (tau + 1 < NT)
+            asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+          else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+        const int u = ks * 4 + (lane / 16);  // 16-B unit 0..7
+        bf16x8 af[4];
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+          const int r = (q * 4 + mi) * 16 + (lane % 16);
+          af[mi] = *reinterpret_cast<const bf16x8*>(
+              ap + (size_t)r * 64 + (size_t)(u ^ (r & 7)) * 8);
+        }
+        if (q == 0) {
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni) {
+            const int r = (wn & 1) * 64 + ni * 16 + (lane % 16);
+            bfr[ni] = *reinterpret_cast<const bf16x8*>(
+                bp + (size_t)r * 64 + (size_t)(u ^ (r & 7)) * 8);
+          }
+        }
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni)
+            acc[q * 4 + mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[mi], bfr[ni], acc[q * 4 + mi][ni], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_barrier();
+      }
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
 #pragma unroll
   for (int mi = 0; mi < 8; ++mi) {
 #pragma unroll
@@ -452,6 +594,24 @@ void launch_gemm(const u16* A, const u16* W, u16* C, const u16* res, int M,
     return;
   }
   const int mt = (M + 255) / 256, nt = (N + 255) / 256;
+  if (var == 3 && (long)mt * nt >= 200 && K % 64 == 0) {
+    static bool attr3 = false;
+    if (!attr3) {
+      hipFuncSetAttribute((const void*)&k_gemm_256p<0>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+      hipFuncSetAttribute((const void*)&k_gemm_256p<1>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+      attr3 = true;
+    }
+    dim3 grid(mt, nt);
+    if (epi == 0)
+      hipLaunchKernelGGL(k_gemm_256p<0>, grid, dim3(512), 131072, s, A, W, C,
+                         res, M, N, K);
+    else
+      hipLaunchKernelGGL(k_gemm_256p<1>, grid, dim3(512), 131072, s, A, W, C,
+                         res, M, N, K);
+    return;
+  }
   if (var >= 1 && (long)mt * nt >= 200) {
     static bool attr_set = false;
     if (!attr_set) {
