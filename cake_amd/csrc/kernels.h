@@ -29,6 +29,7 @@ void launch_gemv_gateup(const u16* W, const u16* x, u16* out, const u16* nw,
 void launch_gemv_res_splitk(const u16* W, const u16* x, u16* out,
                             const u16* res, float* ws, u32* cnt, int N,
                             int K, hipStream_t s);
+// fused rms_norm -> qkv GEMV -> rope -> KV-cache store (llama family)
 void launch_gemv_qkv_rope(const u16* W, const u16* x, u16* out, const u16* nw,
                           float eps, u16* kc, u16* vc, u16* vtc,
                           const float* cost, const float* sint,
