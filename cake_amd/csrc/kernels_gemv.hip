@@ -942,13 +942,28 @@ __global__ __launch_bounds__(256) void k_gemv_gateup_fp8(
 __global__ void k_dequant_fp8(const unsigned char* __restrict__ W,
                               const float* __restrict__ sc,
                               u16* __restrict__ out, int N, int K, int nkb) {
-  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-  size_t stride = (size_t)gridDim.x * blockDim.x;
-  size_t total = (size_t)N * K;
-  for (; i < total; i += stride) {
-    int r = (int)(i / K), k = (int)(i % K);
-    f32x2 v = __builtin_amdgcn_cvt_pk_f32_fp8((u32)W[i], false);
-    out[i] = f2b(v[0] * sc[(size_t)(r / 128) * nkb + k / 128]);
+  // 16 consecutive elements per thread iteration: one 16-B weight load,
+  // one scale (the 128-wide block covers the whole group; K % 16 == 0 on
+  // the fp8 path), one 16-float decode, two 16-B stores.  The scalar
+  // per-byte version measured 2.1 TB/s; this is a plain streaming copy.
+  size_t i16 = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  const size_t total = ((size_t)N * K) / 16;
+  for (; i16 < total; i16 += stride) {
+    const size_t i = i16 * 16;
+    const int r = (int)(i / K), k = (int)(i % K);
+    const uint4v w = *reinterpret_cast<const uint4v*>(W + i);
+    const float s = sc[(size_t)(r / 128) * nkb + k / 128];
+    float wd[16];
+    f8x16_decode(w, wd);
+    short8 o0, o1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      o0[j] = (short)f2b(wd[j] * s);
+      o1[j] = (short)f2b(wd[8 + j] * s);
+    }
+    *reinterpret_cast<short8*>(out + i) = o0;
+    *reinterpret_cast<short8*>(out + i + 8) = o1;
   }
 }
 
