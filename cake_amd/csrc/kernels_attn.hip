@@ -441,23 +441,38 @@ __global__ __launch_bounds__(512) void k_attn_prefill_mfma(
   const u16* kbase = kc + (size_t)kvh * max_seq * hd;
   const u16* vtbase = vtc + (size_t)kvh * hd * max_seq;
 
+  // K fragments for the CURRENT tile live in kf and are reloaded IN PLACE
+  // for tile+1 right after the QK^T MFMAs consume them — the reload and
+  // the current tile's V^T loads (issued before the softmax, in consumer
+  // order) ride the memory system under the softmax VALU chain, the same
+  // wait-at-first-consumer structure as the decode kernel.
+  bf16x8 kf[8];
+  auto load_k = [&](int pkv) {
+    const u16* krow = kbase + (size_t)(pkv + lq) * hd;
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk)
+      kf[kk] = *reinterpret_cast<const bf16x8*>(krow + kk * 16 + lhalf * 8);
+  };
+  load_k(0);
   for (int tile = 0; tile < ntiles; ++tile) {
     const int pkv = tile * 32;
-    // K fragments: A[i=kv][k], lane holds kv = lq
-    bf16x8 kf[8];
-    {
-      const u16* krow = kbase + (size_t)(pkv + lq) * hd;
-#pragma unroll
-      for (int kk = 0; kk < 8; ++kk)
-        kf[kk] =
-            *reinterpret_cast<const bf16x8*>(krow + kk * 16 + lhalf * 8);
-    }
     f32x16 p;
 #pragma unroll
     for (int r = 0; r < 16; ++r) p[r] = 0.f;
 #pragma unroll
     for (int kk = 0; kk < 8; ++kk)
       p = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[kk], qf[kk], p, 0, 0, 0);
+    // current tile's V^T fragments (addresses only — no score dependency)
+    bf16x8 vf[2][4];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int db = 0; db < 4; ++db)
+        vf[kk][db] = *reinterpret_cast<const bf16x8*>(
+            vtbase + (size_t)(db * 32 + lq) * max_seq + pkv + 16 * kk +
+            lhalf * 8);
+    // next tile's K, overwriting kf (the MFMAs above already read it)
+    if (tile + 1 < ntiles) load_k(pkv + 32);
     // scale + causal mask (reg r -> kv row (r&3)+8*(r>>2)+4*lhalf)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -514,15 +529,9 @@ __global__ __launch_bounds__(512) void k_attn_prefill_mfma(
         af.u[3] = pk[4 * kk + 3];
       }
 #pragma unroll
-      for (int db = 0; db < 4; ++db) {
-        const int d = db * 32 + lq;
-        const u16* vt =
-            vtbase + (size_t)d * max_seq + pkv + 16 * kk + lhalf * 8;
-        bf16x8 vf = *reinterpret_cast<const bf16x8*>(vt);
-        oacc[db] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(af.v, vf, oacc[db], 0,
-                                                    0, 0);
-      }
+      for (int db = 0; db < 4; ++db)
+        oacc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            af.v, vf[kk][db], oacc[db], 0, 0, 0);
     }
   }
 
