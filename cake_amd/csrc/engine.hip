@@ -739,6 +739,7 @@ extern "C" void cake_hip_engine_free(cake_engine* e) {
   if (e->wscratch) hipFree(e->wscratch);
   hipFree(e->ids); hipFree(e->ring); hipFree(e->pval); hipFree(e->pidx);
   hipFree(e->attn_ws); hipFree(e->attn_cnt);
+  hipFree(e->gemv_ws); hipFree(e->gemv_cnt);
   hipFree(e->dev_pos); hipFree(e->dev_step);
   hipFree(e->dev_tok); hipFree(e->cos_t); hipFree(e->sin_t);
   hipStreamDestroy(e->stream);
