@@ -73,6 +73,9 @@ extern "C" const char* cake_hip_last_error(void) { return g_err.c_str(); }
 struct ModelConfig {
   int hidden = 0, inter = 0, vocab = 0, layers = 0, nh = 0, nkv = 0;
   int head_dim = 0, max_pos = 4096;
+  int window = 0;  // sliding_window tokens, 0 = full attention
+                   // (cache.rs:173-205 trims KV to the window; here the
+                   // attention span is bounded instead — same semantics)
   float rms_eps = 1e-5f, rope_theta = 10000.f;
   bool tied = false, qk_norm = false;
   bool fp8 = false;  // quantization_config.quant_method == "fp8" (fp8.rs:20-40)
@@ -113,6 +116,11 @@ static int parse_config(const char* json, ModelConfig* c) {
   if (auto p = v->get("tie_word_embeddings")) c->tied = p->bool_or(false);
   if (auto p = v->get("model_type"))
     c->qk_norm = p->str.find("qwen3") != std::string::npos;
+  // sliding window (mistral-style; qwen gates it behind use_sliding_window)
+  if (auto p = v->get("sliding_window"))
+    if (p->kind == minijson::Value::Num) c->window = (int)p->num;
+  if (auto p = v->get("use_sliding_window"))
+    if (!p->bool_or(true)) c->window = 0;
   if (auto qc = v->get("quantization_config"))
     if (auto qm = qc->get("quant_method"))
       c->fp8 = qm->str == "fp8";
@@ -393,7 +401,7 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
                  4.0 * (e->host_pos + 1) * Sq);
     launch_attn_decode(e->qkv, l.kc, l.vc, e->dev_pos, e->attn_ws,
                        e->attn_cnt, e->attn_out, c.nh, c.nkv, hd, e->max_seq,
-                       e->nchunk, e->stream);
+                       e->nchunk, c.window, e->stream);
   }
   {  // o projection + residual
     double wb = (double)H * Sq * (c.fp8 ? 1 : 2);
@@ -477,7 +485,8 @@ static void enqueue_layer_prefill(cake_engine* e, LayerDev& l, int S,
     StatScope ss(e, "attn_prefill", 2.0 * S * n_avg * 2 * hd * c.nh / 4,
                  4.0 * S * n_avg * hd * c.nh);
     launch_attn_prefill(e->qkv, l.kc, l.vc, l.vtc, e->attn_out, S, pos0,
-                        c.nh, c.nkv, hd, e->max_seq, Nq, Sq, e->stream);
+                        c.nh, c.nkv, hd, e->max_seq, Nq, Sq, c.window,
+                        e->stream);
   }
   if (c.fp8) {
     StatScope ss(e, "dequant_fp8", (double)H * Sq * 3, 0);

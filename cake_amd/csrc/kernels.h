@@ -61,12 +61,12 @@ void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, u16* vtc,
                                const u16* kn, float eps, hipStream_t s);
 void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
                         const int* pos, float* ws, u32* cnt, u16* out, int nh,
-                        int nkv, int hd, int max_seq, int nchunk,
+                        int nkv, int hd, int max_seq, int nchunk, int window,
                         hipStream_t s);
 void launch_attn_prefill(const u16* qkv, const u16* kc, const u16* vc,
                          const u16* vtc, u16* out, int S, int pos0, int nh,
                          int nkv, int hd, int max_seq, int qkv_stride,
-                         int out_stride, hipStream_t s);
+                         int out_stride, int window, hipStream_t s);
 void launch_rope_simple(u16* x, const float* cost, const float* sint, int bh,
                         int s, int d, hipStream_t st);
 void launch_argmax(const float* logits, int n, float* pval, int* pidx,

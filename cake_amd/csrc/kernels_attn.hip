@@ -141,12 +141,15 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
     const u16* __restrict__ kc, const u16* __restrict__ vc,
     const int* __restrict__ pos, float* __restrict__ ws,
     u32* __restrict__ cnt, u16* __restrict__ outbuf, int nh, int nkv,
-    int hd, int max_seq, int nchunk) {
+    int hd, int max_seq, int nchunk, int window) {
   const int h = blockIdx.y;
   const int chunk = blockIdx.x;
   const int n = *pos + 1;
-  const int cs = (n + nchunk - 1) / nchunk;
-  const int start = chunk * cs;
+  // sliding window (cache.rs:173-205 semantics): attention spans only the
+  // last `window` positions; chunks partition [lo, n)
+  const int lo = (window > 0 && n > window) ? n - window : 0;
+  const int cs = (n - lo + nchunk - 1) / nchunk;
+  const int start = lo + chunk * cs;
   const int end = min(start + cs, n);
   const int kvh = h / (nh / nkv);
   const int t = threadIdx.x, wid = t / WAVE, lane = t % WAVE;
@@ -340,7 +343,8 @@ __global__ __launch_bounds__(256) void k_attn_decode_fused(
 __global__ __launch_bounds__(256) void k_attn_prefill(
     const u16* __restrict__ qkv, const u16* __restrict__ kc,
     const u16* __restrict__ vc, u16* __restrict__ out, int S, int pos0,
-    int nh, int nkv, int hd, int max_seq, int qkv_stride, int out_stride) {
+    int nh, int nkv, int hd, int max_seq, int qkv_stride, int out_stride,
+    int window) {
   const int h = blockIdx.y;
   const int t = threadIdx.x, wid = t / WAVE, lane = t % WAVE;
   const int sidx = blockIdx.x * 4 + wid;
@@ -359,7 +363,8 @@ __global__ __launch_bounds__(256) void k_attn_prefill(
   float m = -INFINITY, l = 0.f, o0 = 0.f, o1 = 0.f;
   const u16* kbase = kc + (size_t)kvh * max_seq * hd;
   const u16* vbase = vc + (size_t)kvh * max_seq * hd;
-  for (int p = 0; p < n; ++p) {
+  const int p0 = (window > 0 && n > window) ? n - window : 0;
+  for (int p = p0; p < n; ++p) {
     float dot = 0.f;
     if (act) {
       const u16* kr = kbase + (size_t)p * hd + e0;
@@ -406,7 +411,8 @@ __global__ __launch_bounds__(256) void k_attn_prefill(
 __global__ __launch_bounds__(512) void k_attn_prefill_mfma(
     const u16* __restrict__ qkv, const u16* __restrict__ kc,
     const u16* __restrict__ vtc, u16* __restrict__ out, int S, int pos0,
-    int nh, int nkv, int max_seq, int qkv_stride, int out_stride) {
+    int nh, int nkv, int max_seq, int qkv_stride, int out_stride,
+    int window) {
   const int hd = 128;
   const int w = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
@@ -438,6 +444,10 @@ __global__ __launch_bounds__(512) void k_attn_prefill_mfma(
   const bool q_valid = qb + lq < S;
   const int n_wave = pos0 + min(qb + 32, S);  // kv needed by this wave
   const int ntiles = (n_wave + 31) / 32;
+  // sliding window: the wave's SMALLEST query attends from
+  // (pos0 + qb) - window + 1; whole tiles below that are skipped
+  const int tile0 =
+      (window > 0) ? max(0, pos0 + qb - (window - 1)) / 32 : 0;
   const u16* kbase = kc + (size_t)kvh * max_seq * hd;
   const u16* vtbase = vtc + (size_t)kvh * hd * max_seq;
 
@@ -453,8 +463,8 @@ __global__ __launch_bounds__(512) void k_attn_prefill_mfma(
     for (int kk = 0; kk < 8; ++kk)
       kf[kk] = *reinterpret_cast<const bf16x8*>(krow + kk * 16 + lhalf * 8);
   };
-  load_k(0);
-  for (int tile = 0; tile < ntiles; ++tile) {
+  load_k(tile0 * 32);
+  for (int tile = tile0; tile < ntiles; ++tile) {
     const int pkv = tile * 32;
     f32x16 p;
 #pragma unroll
@@ -477,7 +487,9 @@ __global__ __launch_bounds__(512) void k_attn_prefill_mfma(
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kv_abs = pkv + (r & 3) + 8 * (r >> 2) + 4 * lhalf;
-      p[r] = (q_valid && kv_abs <= q_abs) ? p[r] * scale : -INFINITY;
+      const bool vis = q_valid && kv_abs <= q_abs &&
+                       (window == 0 || kv_abs > q_abs - window);
+      p[r] = vis ? p[r] * scale : -INFINITY;
     }
     // online softmax stats for column q = lq (halves combined via xor-32)
     float tm = -INFINITY;
@@ -575,23 +587,23 @@ void launch_rope_store_prefill(u16* qkv, u16* kc, u16* vc, u16* vtc,
 }
 void launch_attn_decode(const u16* q, const u16* kc, const u16* vc,
                         const int* pos, float* ws, u32* cnt, u16* out, int nh,
-                        int nkv, int hd, int max_seq, int nchunk,
+                        int nkv, int hd, int max_seq, int nchunk, int window,
                         hipStream_t s) {
   hipLaunchKernelGGL(k_attn_decode_fused, dim3(nchunk, nh), dim3(256), 0, s,
                      q, kc, vc, pos, ws, cnt, out, nh, nkv, hd, max_seq,
-                     nchunk);
+                     nchunk, window);
 }
 void launch_attn_prefill(const u16* qkv, const u16* kc, const u16* vc,
                          const u16* vtc, u16* out, int S, int pos0, int nh,
                          int nkv, int hd, int max_seq, int qkv_stride,
-                         int out_stride, hipStream_t s) {
+                         int out_stride, int window, hipStream_t s) {
   if (hd == 128) {
     hipLaunchKernelGGL(k_attn_prefill_mfma, dim3((S + 255) / 256, nh),
                        dim3(512), 0, s, qkv, kc, vtc, out, S, pos0, nh, nkv,
-                       max_seq, qkv_stride, out_stride);
+                       max_seq, qkv_stride, out_stride, window);
   } else {
     hipLaunchKernelGGL(k_attn_prefill, dim3((S + 3) / 4, nh), dim3(256), 0, s,
                        qkv, kc, vc, out, S, pos0, nh, nkv, hd, max_seq,
-                       qkv_stride, out_stride);
+                       qkv_stride, out_stride, window);
   }
 }

@@ -78,6 +78,9 @@ class Config:
     # llama3 rope scaling (config.rs:50-65); None = no scaling
     rope_scaling: Optional[dict] = None
     model_prefix: str = "model"             # config.rs:104
+    # sliding-window attention span (cache.rs:173-205 trims KV to the
+    # window; bounding the attention span is equivalent); None = full
+    sliding_window: Optional[int] = None
 
     @property
     def hd(self) -> int:
@@ -111,6 +114,8 @@ class Config:
             tie_word_embeddings=j.get("tie_word_embeddings", False),
             use_qk_norm="qwen3" in j.get("model_type", ""),
             rope_scaling=rs,
+            sliding_window=(j.get("sliding_window")
+                            if j.get("use_sliding_window", True) else None),
         )
 
 
@@ -196,17 +201,21 @@ def rope(x: np.ndarray, cos: np.ndarray, sin: np.ndarray) -> np.ndarray:
     return np.concatenate([x1 * c - x2 * s, x2 * c + x1 * s], axis=-1)
 
 
-def causal_mask(seq_len: int, kv_len: int) -> np.ndarray:
+def causal_mask(seq_len: int, kv_len: int,
+                window: Optional[int] = None) -> np.ndarray:
     """cache.rs:150-160 + attention.rs:324-333: True where masked.
 
-    Square (seq,seq) mask with j > i masked, left-padded with zeros (attend)
-    when kv_len > seq_len (prefill into an existing cache).
-    """
-    sq = np.triu(np.ones((seq_len, seq_len), dtype=bool), k=1)
-    if kv_len > seq_len:
-        pad = np.zeros((seq_len, kv_len - seq_len), dtype=bool)
-        return np.concatenate([pad, sq], axis=1)
-    return sq
+    Causal (j > pos_i) over absolute positions pos_i = kv_len - seq + i,
+    left-padded with zeros (attend) when kv_len > seq_len (prefill into an
+    existing cache).  With a sliding window W, positions j <= pos_i - W are
+    masked too (cache.rs:173-205 trims the KV to the window; masking the
+    span is equivalent)."""
+    pos = kv_len - seq_len + np.arange(seq_len)[:, None]
+    j = np.arange(kv_len)[None, :]
+    m = j > pos
+    if window:
+        m = m | (j <= pos - window)
+    return m
 
 
 # ---------------------------------------------------------------------------
@@ -297,8 +306,8 @@ class OracleModel:
         vr = np.repeat(v, rep, axis=1)
         # f32 attention (attention.rs:300-343)
         att = q @ kr.transpose(0, 1, 3, 2) / np.float32(np.sqrt(hd))
-        if s > 1:
-            m = causal_mask(s, kv_len)
+        if s > 1 or self.cfg.sliding_window:
+            m = causal_mask(s, kv_len, self.cfg.sliding_window)
             att = np.where(m[None, None], np.float32(-np.inf), att)
         att = softmax_lastdim(att)
         y = att @ vr

@@ -40,6 +40,19 @@ def tiny_llama_cfg():
     )
 
 
+def tiny_mistral_cfg():
+    """Mistral shaped (sliding-window attention, plain rope), tiny dims.
+    The prompt (48 tokens) exceeds the window (24) so the windowed mask is
+    actually exercised (cache.rs:173-205 trim semantics)."""
+    return dict(
+        model_type="mistral", hidden_size=64, intermediate_size=128,
+        vocab_size=256, num_hidden_layers=2, num_attention_heads=4,
+        num_key_value_heads=2, head_dim=16, rms_norm_eps=1e-5,
+        rope_theta=10000.0, max_position_embeddings=256,
+        tie_word_embeddings=False, sliding_window=24,
+    )
+
+
 def tiny_qwen3_cfg():
     """Qwen3 shaped (qk_norm, tied embeddings), tiny dims."""
     return dict(
@@ -72,6 +85,23 @@ def hf_model(cfg_json):
             attention_bias=False, mlp_bias=False,
         )
         m = transformers.LlamaForCausalLM(cfg)
+    elif cfg_json["model_type"] == "mistral":
+        cfg = transformers.MistralConfig(
+            hidden_size=cfg_json["hidden_size"],
+            intermediate_size=cfg_json["intermediate_size"],
+            vocab_size=cfg_json["vocab_size"],
+            num_hidden_layers=cfg_json["num_hidden_layers"],
+            num_attention_heads=cfg_json["num_attention_heads"],
+            num_key_value_heads=cfg_json["num_key_value_heads"],
+            head_dim=cfg_json["head_dim"],
+            rms_norm_eps=cfg_json["rms_norm_eps"],
+            rope_theta=cfg_json["rope_theta"],
+            max_position_embeddings=cfg_json["max_position_embeddings"],
+            tie_word_embeddings=cfg_json["tie_word_embeddings"],
+            sliding_window=cfg_json["sliding_window"],
+            attn_implementation="eager",
+        )
+        m = transformers.MistralForCausalLM(cfg)
     else:
         cfg = transformers.Qwen3Config(
             hidden_size=cfg_json["hidden_size"],
@@ -193,6 +223,7 @@ def run_one(name, cfg_json, prompt_len=17, gen=12):
 def main():
     run_one("tiny_llama3", tiny_llama_cfg())
     run_one("tiny_qwen3", tiny_qwen3_cfg())
+    run_one("tiny_mistral", tiny_mistral_cfg(), prompt_len=48)
 
 
 if __name__ == "__main__":
