@@ -8,8 +8,11 @@ Mirrors cake's API surface (cake-core/src/cake/sharding/api/mod.rs:66-110):
 
 Generation is the same hot loop the bench measures (Master::generate_text,
 master.rs:109-171): greedy ArgMax when temperature <= 0, on-GPU
-Gumbel-argmax sampling otherwise (cake's own temperature-sampling trick,
-text_model.rs:102-118); top-k/top-p are round-2 items.
+Gumbel-argmax sampling for plain temperature (cake's own
+temperature-sampling trick, text_model.rs:102-118).  top_k/top_p requests
+take a host-sampling loop over per-token logits — faithfully mirroring
+cake, whose LogitsProcessor (candle Sampling::TopK/TopP/TopKThenTopP,
+created at text_model.rs:102-118) also samples on the host.
 
 Tokenization: pass a `tokenizers.Tokenizer` (tokenizer.json) for text
 prompts; without one, requests supply `prompt_token_ids` directly (the
@@ -36,6 +39,28 @@ def build_prompt(messages):
              for m in messages]
     parts.append("assistant:")
     return "\n".join(parts)
+
+
+def sample_from_logits(logits, temperature, top_k, top_p, rng):
+    """Restates candle's Sampling::{TopK,TopP,TopKThenTopP,All} as cake
+    configures it (text_model.rs:102-118): softmax at `temperature`, keep
+    the top-k, then the minimal nucleus reaching top_p (always >= 1
+    token), renormalize, multinomial draw."""
+    x = np.asarray(logits, dtype=np.float64) / max(1e-6, temperature)
+    x -= x.max()
+    p = np.exp(x)
+    p /= p.sum()
+    order = np.argsort(-p)
+    if top_k:
+        order = order[:int(top_k)]
+    kept = p[order]
+    if top_p and top_p < 1.0:
+        c = np.cumsum(kept) / kept.sum()
+        cut = int(np.searchsorted(c, top_p)) + 1
+        order = order[:cut]
+        kept = kept[:cut]
+    kept = kept / kept.sum()
+    return int(rng.choice(order, p=kept))
 
 
 class GenSession:
@@ -74,6 +99,26 @@ class GenSession:
                 if produced >= max_tokens or t in self.eos_ids:
                     return
 
+    def generate_sampled(self, prompt_ids, max_tokens, temperature, top_k,
+                         top_p, seed):
+        """Host-sampling loop for top-k/top-p: per-token logits from
+        append-prefill steps, sampled like cake's host-side
+        LogitsProcessor (slower than the graph-replayed decode loop, same
+        trade cake makes)."""
+        self.engine.reset()
+        rng = np.random.default_rng(seed)
+        _, logits = self.engine.prefill(
+            np.asarray(prompt_ids, dtype=np.uint32), want_logits=True)
+        produced = 0
+        while True:
+            tok = sample_from_logits(logits, temperature, top_k, top_p, rng)
+            yield tok
+            produced += 1
+            if produced >= max_tokens or tok in self.eos_ids:
+                return
+            _, logits = self.engine.prefill(
+                np.asarray([tok], dtype=np.uint32), want_logits=True)
+
 
 def create_app(engine, model_name="cake-amd", tokenizer=None, eos_ids=(),
                topology=None):
@@ -107,9 +152,14 @@ def create_app(engine, model_name="cake-amd", tokenizer=None, eos_ids=(),
     async def run_request(body, kind):
         max_tokens = int(body.get("max_tokens", 128))
         stream = bool(body.get("stream", False))
-        if hasattr(engine, "set_sampling"):
-            engine.set_sampling(float(body.get("temperature", 0.0)),
-                                int(body.get("seed", 299792458)))
+        temperature = float(body.get("temperature", 0.0))
+        top_k = body.get("top_k")
+        top_p = body.get("top_p")
+        seed = int(body.get("seed", 299792458))
+        use_host_sampling = temperature > 0 and (top_k or
+                                                 (top_p and top_p < 1.0))
+        if not use_host_sampling and hasattr(engine, "set_sampling"):
+            engine.set_sampling(temperature, seed)
         if "prompt_token_ids" in body:
             ids = [int(t) for t in body["prompt_token_ids"]]
         elif kind == "chat":
@@ -119,9 +169,15 @@ def create_app(engine, model_name="cake-amd", tokenizer=None, eos_ids=(),
         rid = f"chatcmpl-{uuid.uuid4().hex[:12]}"
         created = int(time.time())
 
+        def gen_stream():
+            if use_host_sampling:
+                return sess.generate_sampled(ids, max_tokens, temperature,
+                                             top_k, top_p, seed)
+            return sess.generate(ids, max_tokens)
+
         if stream:
             def sse():
-                for tok in sess.generate(ids, max_tokens):
+                for tok in gen_stream():
                     delta = ({"content": decode_tok(tok)} if kind == "chat"
                              else None)
                     chunk = {
@@ -140,7 +196,7 @@ def create_app(engine, model_name="cake-amd", tokenizer=None, eos_ids=(),
                 yield "data: [DONE]\n\n"
             return StreamingResponse(sse(), media_type="text/event-stream")
 
-        toks = list(sess.generate(ids, max_tokens))
+        toks = list(gen_stream())
         text = ("".join(decode_tok(t) for t in toks) if tokenizer
                 else None)
         finish = "stop" if (toks and toks[-1] in sess.eos_ids) else "length"
