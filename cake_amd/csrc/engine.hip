@@ -76,6 +76,12 @@ struct ModelConfig {
   int window = 0;  // sliding_window tokens, 0 = full attention
                    // (cache.rs:173-205 trims KV to the window; here the
                    // attention span is bounded instead — same semantics)
+  int mwl = 0;     // max_window_layers: layers < mwl attend FULL (qwen
+                   // rule; HF layer_types condenses to this for the
+                   // full-then-sliding pattern); 0 = all layers windowed
+  int win_for(int layer_idx) const {
+    return (window > 0 && layer_idx >= mwl) ? window : 0;
+  }
   float rms_eps = 1e-5f, rope_theta = 10000.f;
   bool tied = false, qk_norm = false;
   bool fp8 = false;  // quantization_config.quant_method == "fp8" (fp8.rs:20-40)
@@ -116,11 +122,28 @@ static int parse_config(const char* json, ModelConfig* c) {
   if (auto p = v->get("tie_word_embeddings")) c->tied = p->bool_or(false);
   if (auto p = v->get("model_type"))
     c->qk_norm = p->str.find("qwen3") != std::string::npos;
-  // sliding window (mistral-style; qwen gates it behind use_sliding_window)
+  // sliding window (mistral-style; qwen gates it behind use_sliding_window
+  // and applies it only from max_window_layers up; HF "layer_types" has
+  // the same full-then-sliding shape for these families)
   if (auto p = v->get("sliding_window"))
     if (p->kind == minijson::Value::Num) c->window = (int)p->num;
   if (auto p = v->get("use_sliding_window"))
     if (!p->bool_or(true)) c->window = 0;
+  if (auto p = v->get("max_window_layers"))
+    if (p->kind == minijson::Value::Num) c->mwl = (int)p->num;
+  if (auto lt = v->get("layer_types"))
+    if (lt->kind == minijson::Value::Arr) {
+      // condense to the full-then-sliding boundary; reject other patterns
+      int first_sliding = -1;
+      for (size_t i = 0; i < lt->arr.size(); ++i) {
+        const bool sl = lt->arr[i]->str == "sliding_attention";
+        if (sl && first_sliding < 0) first_sliding = (int)i;
+        if (!sl && first_sliding >= 0)
+          return set_err(5, "layer_types: only the full-then-sliding "
+                            "pattern is supported");
+      }
+      c->mwl = first_sliding < 0 ? c->layers : first_sliding;
+    }
   if (auto qc = v->get("quantization_config"))
     if (auto qm = qc->get("quant_method"))
       c->fp8 = qm->str == "fp8";
@@ -206,6 +229,7 @@ struct Stats {
 // engine
 // ---------------------------------------------------------------------------
 struct LayerDev {
+  int idx = 0;  // absolute layer index (per-layer attention window)
   u16 *rms1 = nullptr, *rms2 = nullptr;
   u16 *wqkv = nullptr, *wo = nullptr, *wgu = nullptr, *wdown = nullptr;
   u16 *qnorm = nullptr, *knorm = nullptr;
@@ -401,7 +425,7 @@ static void enqueue_layer_decode(cake_engine* e, LayerDev& l) {
                  4.0 * (e->host_pos + 1) * Sq);
     launch_attn_decode(e->qkv, l.kc, l.vc, e->dev_pos, e->attn_ws,
                        e->attn_cnt, e->attn_out, c.nh, c.nkv, hd, e->max_seq,
-                       e->nchunk, c.window, e->stream);
+                       e->nchunk, c.win_for(l.idx), e->stream);
   }
   {  // o projection + residual
     double wb = (double)H * Sq * (c.fp8 ? 1 : 2);
@@ -485,8 +509,8 @@ static void enqueue_layer_prefill(cake_engine* e, LayerDev& l, int S,
     StatScope ss(e, "attn_prefill", 2.0 * S * n_avg * 2 * hd * c.nh / 4,
                  4.0 * S * n_avg * hd * c.nh);
     launch_attn_prefill(e->qkv, l.kc, l.vc, l.vtc, e->attn_out, S, pos0,
-                        c.nh, c.nkv, hd, e->max_seq, Nq, Sq, c.window,
-                        e->stream);
+                        c.nh, c.nkv, hd, e->max_seq, Nq, Sq,
+                        c.win_for(l.idx), e->stream);
   }
   if (c.fp8) {
     StatScope ss(e, "dequant_fp8", (double)H * Sq * 3, 0);
@@ -619,6 +643,8 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
   if (c.fp8 && (H % 128 || I % 128 || Sq % 128 || c.skv() % 128))
     return set_err(5, "fp8 requires dims to be multiples of 128");
   e->L.resize(layer_hi - layer_lo);
+  for (size_t li = 0; li < e->L.size(); ++li)
+    e->L[li].idx = layer_lo + (int)li;
   for (auto& l : e->L) {
     ALLOC(l.rms1, u16, H);
     ALLOC(l.rms2, u16, H);

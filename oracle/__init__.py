@@ -79,8 +79,17 @@ class Config:
     rope_scaling: Optional[dict] = None
     model_prefix: str = "model"             # config.rs:104
     # sliding-window attention span (cache.rs:173-205 trims KV to the
-    # window; bounding the attention span is equivalent); None = full
+    # window; bounding the attention span is equivalent); None = full.
+    # layer_windows: per-layer span (Gemma3/EXAONE-style interleave of
+    # local and global layers; qwen-style max_window_layers) — tuple of
+    # (window-or-None) per layer, built by from_json
     sliding_window: Optional[int] = None
+    layer_windows: Optional[tuple] = None
+
+    def window_for(self, block_idx: int) -> Optional[int]:
+        if self.layer_windows is not None:
+            return self.layer_windows[block_idx]
+        return self.sliding_window
 
     @property
     def hd(self) -> int:
@@ -116,7 +125,26 @@ class Config:
             rope_scaling=rs,
             sliding_window=(j.get("sliding_window")
                             if j.get("use_sliding_window", True) else None),
+            layer_windows=cls._layer_windows(j),
         )
+
+    @staticmethod
+    def _layer_windows(j: dict) -> Optional[tuple]:
+        '''Per-layer attention window.  HF "layer_types" wins
+        ("sliding_attention"/"full_attention" per layer); otherwise the
+        qwen-style rule: sliding for layer_idx >= max_window_layers when
+        use_sliding_window, all layers when only sliding_window is set
+        (mistral).'''
+        w = j.get("sliding_window")
+        if not w or not j.get("use_sliding_window", True):
+            return None
+        L = j["num_hidden_layers"]
+        lt = j.get("layer_types")
+        if lt is not None:
+            return tuple(w if t == "sliding_attention" else None
+                         for t in lt)
+        mwl = j.get("max_window_layers", 0)
+        return tuple(w if i >= mwl else None for i in range(L))
 
 
 # ---------------------------------------------------------------------------
@@ -306,8 +334,9 @@ class OracleModel:
         vr = np.repeat(v, rep, axis=1)
         # f32 attention (attention.rs:300-343)
         att = q @ kr.transpose(0, 1, 3, 2) / np.float32(np.sqrt(hd))
-        if s > 1 or self.cfg.sliding_window:
-            m = causal_mask(s, kv_len, self.cfg.sliding_window)
+        win = self.cfg.window_for(block_idx)
+        if s > 1 or win:
+            m = causal_mask(s, kv_len, win)
             att = np.where(m[None, None], np.float32(-np.inf), att)
         att = softmax_lastdim(att)
         y = att @ vr

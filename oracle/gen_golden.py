@@ -64,6 +64,16 @@ def tiny_qwen3_cfg():
     )
 
 
+def tiny_qwen3_swa_cfg():
+    """Qwen3 with INTERLEAVED sliding-window layers (max_window_layers=1:
+    layer 0 full, layer 1 windowed) — the per-layer window pattern
+    Gemma3/EXAONE-class models use."""
+    cfg = tiny_qwen3_cfg()
+    cfg.update(use_sliding_window=True, sliding_window=24,
+               max_window_layers=1)
+    return cfg
+
+
 def hf_model(cfg_json):
     import torch
     import transformers
@@ -116,6 +126,15 @@ def hf_model(cfg_json):
             max_position_embeddings=cfg_json["max_position_embeddings"],
             tie_word_embeddings=cfg_json["tie_word_embeddings"],
         )
+        if cfg_json.get("use_sliding_window"):
+            cfg.use_sliding_window = True
+            cfg.sliding_window = cfg_json["sliding_window"]
+            cfg.max_window_layers = cfg_json["max_window_layers"]
+            cfg.layer_types = ["full_attention" if i <
+                               cfg_json["max_window_layers"] else
+                               "sliding_attention"
+                               for i in range(cfg_json["num_hidden_layers"])]
+            cfg._attn_implementation = "eager"
         m = transformers.Qwen3ForCausalLM(cfg)
     m.eval()
     m = m.float()
@@ -224,6 +243,7 @@ def main():
     run_one("tiny_llama3", tiny_llama_cfg())
     run_one("tiny_qwen3", tiny_qwen3_cfg())
     run_one("tiny_mistral", tiny_mistral_cfg(), prompt_len=48)
+    run_one("tiny_qwen3swa", tiny_qwen3_swa_cfg(), prompt_len=48)
 
 
 if __name__ == "__main__":

@@ -118,7 +118,7 @@ def test_op_gemm(M, N, K):
 # ---------------------------------------------------------------------------
 # engine vs oracle on the committed golden fixtures
 # ---------------------------------------------------------------------------
-@pytest.fixture(scope="module", params=["tiny_llama3", "tiny_qwen3", "tiny_mistral"])
+@pytest.fixture(scope="module", params=["tiny_llama3", "tiny_qwen3", "tiny_mistral", "tiny_qwen3swa"])
 def fixture_engine(request, tmp_path_factory):
     import os
     golden = os.path.join(os.path.dirname(os.path.abspath(__file__)),
