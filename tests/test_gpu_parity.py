@@ -464,6 +464,47 @@ def test_adaptive_nchunk_recapture():
             eng.close()
 
 
+def test_gqa_ratio8_hd128_parity():
+    """GQA ratio 8 (the 70B/32B head layout: nh/nkv = 8) with hd=128 —
+    exercises the kvh = h/8 mapping in the decode and MFMA-prefill
+    kernels, which the ratio-2 golden fixtures never touch."""
+    import os
+    import tempfile
+    cfg_json = dict(
+        model_type="llama", hidden_size=256, intermediate_size=512,
+        vocab_size=512, num_hidden_layers=2, num_attention_heads=8,
+        num_key_value_heads=1, head_dim=128, rms_norm_eps=1e-5,
+        rope_theta=500000.0, max_position_embeddings=1024,
+        tie_word_embeddings=False)
+    cfg = Config.from_json(cfg_json)
+    w = random_weights(cfg, seed=271)
+    oracle = quantized_oracle(cfg, w)
+    with tempfile.TemporaryDirectory() as td:
+        st = os.path.join(td, "m.safetensors")
+        weights_to_safetensors(w, cfg, st)
+        eng = cake_amd.Engine(json.dumps(cfg_json), max_seq=512,
+                              max_batch_tokens=256)
+        eng.load_safetensors(st)
+        try:
+            rng = np.random.default_rng(17)
+            prompt = rng.integers(0, cfg.vocab_size,
+                                  size=75).astype(np.uint32)
+            first, lg = eng.prefill(prompt, want_logits=True)
+            ref = oracle.forward(prompt[None, :].astype(np.int64), 0)[0]
+            assert rel_err(lg, ref) < 2e-2
+            toks = eng.decode(6)
+            seq = np.concatenate([prompt, [first], toks[:-1]]).astype(
+                np.uint32)
+            eng.reset()
+            _, lg2 = eng.prefill(seq, want_logits=True)
+            assert int(np.argmax(lg2)) == int(toks[-1])
+            oracle.reset()
+            ref2 = oracle.forward(seq[None, :].astype(np.int64), 0)[0]
+            assert rel_err(lg2, ref2) < 2e-2
+        finally:
+            eng.close()
+
+
 def test_max_seq_guard():
     cfg_json = dict(
         model_type="llama", hidden_size=64, intermediate_size=128,
