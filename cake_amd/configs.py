@@ -39,10 +39,22 @@ QWEN3_32B_FP8 = dict(QWEN3_32B,
     quantization_config=dict(quant_method="fp8",
                              weight_block_size=[128, 128]))
 
+# Mistral-7B-v0.1 shape (mistral/, 196 LoC of the same common Transformer +
+# the sliding-window flag cache.rs:173-205 consumes; window now runs in the
+# attention kernels' span bound)
+MISTRAL_7B = dict(
+    model_type="mistral", hidden_size=4096, intermediate_size=14336,
+    vocab_size=32000, num_hidden_layers=32, num_attention_heads=32,
+    num_key_value_heads=8, head_dim=128, rms_norm_eps=1e-5,
+    rope_theta=10000.0, max_position_embeddings=32768,
+    tie_word_embeddings=False, sliding_window=4096,
+)
+
 MODELS = {
     "llama3-8b": LLAMA3_8B,
     "llama3-70b": LLAMA3_70B,
     "qwen3-0.6b": QWEN3_0_6B,
+    "mistral-7b": MISTRAL_7B,
     "qwen3-32b": QWEN3_32B,
     "qwen3-32b-fp8": QWEN3_32B_FP8,
 }
