@@ -620,6 +620,8 @@ extern "C" int cake_hip_engine_create(const char* config_json, int layer_lo,
                    c.hd());
   if (c.hidden > 16384)
     return set_err(5, "hidden_size %d unsupported (> 16384)", c.hidden);
+  if (c.hidden % 8 || c.inter % 8)
+    return set_err(5, "hidden/intermediate must be multiples of 8");
   if (max_seq <= 0) max_seq = c.max_pos;
   max_seq = (max_seq + 31) & ~31;  // MFMA prefill reads whole 32-pos tiles
   if (max_batch_tokens <= 0) max_batch_tokens = 2048;
@@ -1493,6 +1495,9 @@ extern "C" int cake_hip_op_rms_norm(int rows, int cols, float eps,
 
 extern "C" int cake_hip_op_linear(int M, int N, int K, const float* x,
                                   const float* w, float* out, int device) {
+  // the bf16x8 load granularity of every linear kernel (GEMV and GEMM)
+  if (K < 8 || K % 8 != 0)
+    return set_err(5, "op_linear: K must be a multiple of 8 (got %d)", K);
   OpCtx o;
   size_t n = std::max((size_t)M * K, std::max((size_t)N * K, (size_t)M * N));
   int r = op_ctx(device, n, &o);
@@ -1502,8 +1507,8 @@ extern "C" int cake_hip_op_linear(int M, int N, int K, const float* x,
   if (M == 1) {
     launch_gemv(o.b, o.a, o.c, nullptr, nullptr, 0.f, N, K, 0, o.s);
   } else {
-    if (K % 64)
-      return set_err(5, "gemm requires K %% 64 == 0 (got %d)", K);
+    // ragged K (% 64 != 0, % 8 == 0) handled by the 128^2 kernel's masked
+    // tail tile; the dispatcher keeps such shapes off the big-tile paths
     launch_gemm(o.a, o.b, o.c, nullptr, M, N, K, 0, o.s);
   }
   return down16(o.c, o.fc, out, (size_t)M * N, o.s);

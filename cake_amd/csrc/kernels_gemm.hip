@@ -53,9 +53,12 @@ __global__ __launch_bounds__(256) void k_gemm_bf16(
     for (int j = 0; j < 4; ++j) {
       const int r = (wid * 4 + j) * 8 + lane / 8;
       const int u = (lane % 8) ^ (r & 7);
+      // ragged-K tail: clamp the column so the last tile's loads stay in
+      // the row (the compute loop zeroes the A fragments beyond K)
+      const int kcol = min(ktbase + u * 8, K - 8);
       {  // A tile
         int grow = min(m0 + r, M - 1);
-        const u16* src = A + (size_t)grow * K + ktbase + u * 8;
+        const u16* src = A + (size_t)grow * K + kcol;
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)(uintptr_t)src,
             (__attribute__((address_space(3))) void*)(uintptr_t)(
@@ -64,7 +67,7 @@ __global__ __launch_bounds__(256) void k_gemm_bf16(
       }
       {  // W tile (B^T)
         int grow = min(n0 + r, N - 1);
-        const u16* src = W + (size_t)grow * K + ktbase + u * 8;
+        const u16* src = W + (size_t)grow * K + kcol;
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)(uintptr_t)src,
             (__attribute__((address_space(3))) void*)(uintptr_t)(
@@ -80,7 +83,11 @@ __global__ __launch_bounds__(256) void k_gemm_bf16(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  const int ntiles = K / GEMM_BK;
+  // ragged K (K % 64 != 0, K % 8 == 0 contract): the LAST tile's stage
+  // clamps its loads into the rows and the A fragments beyond K are
+  // zeroed here, so the clamped duplicates contribute exact zeros
+  const int ntiles = (K + GEMM_BK - 1) / GEMM_BK;
+  const bool ragged = (K % GEMM_BK) != 0;
   stage(0, 0);
   __syncthreads();  // drains the glds (vmcnt(0) implied by the barrier)
 
@@ -89,6 +96,7 @@ __global__ __launch_bounds__(256) void k_gemm_bf16(
     if (kt + 1 < ntiles) stage(cur ^ 1, kt + 1);
     const u16* as = As + (size_t)cur * GEMM_BM * GEMM_BK;
     const u16* bs = Bs + (size_t)cur * GEMM_BN * GEMM_BK;
+    const bool tail = ragged && kt == ntiles - 1;
 #pragma unroll
     for (int ks = 0; ks < GEMM_BK / 32; ++ks) {
       bf16x8 af[4], bf[4];
@@ -102,6 +110,11 @@ __global__ __launch_bounds__(256) void k_gemm_bf16(
         const int br = wc * 64 + i * 16 + (lane % 16);
         bf[i] = *reinterpret_cast<const bf16x8*>(
             bs + (size_t)br * GEMM_BK + (size_t)(ku ^ (br & 7)) * 8);
+      }
+      if (tail && kt * GEMM_BK + kk >= K) {
+        const bf16x8 zb = {};
+#pragma unroll
+        for (int i = 0; i < 4; ++i) af[i] = zb;
       }
 #pragma unroll
       for (int i = 0; i < 4; ++i)
@@ -612,7 +625,7 @@ void launch_gemm(const u16* A, const u16* W, u16* C, const u16* res, int M,
                          res, M, N, K);
     return;
   }
-  if (var >= 1 && (long)mt * nt >= 200) {
+  if (var >= 1 && (long)mt * nt >= 200 && K % 64 == 0) {
     static bool attr_set = false;
     if (!attr_set) {
       hipFuncSetAttribute((const void*)&k_gemm_256<0>,

@@ -103,6 +103,8 @@ def test_op_gemv_small_k():
     (33, 128256 // 16, 64),  # wide-N-ish
     (2048, 28672, 4096),   # gate_up prefill shape (256^2 counted-vmcnt path)
     (300, 51300, 256),     # 256^2 path with M and N tails
+    (256, 384, 96),        # ragged K (% 64 != 0): masked tail tile
+    (200, 300, 40),        # ragged K smaller than one 64-wide tile
 ])
 def test_op_gemm(M, N, K):
     # asymmetric random operands (transpose-detecting — guide §5.4 rule 16)
