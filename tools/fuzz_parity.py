@@ -60,8 +60,9 @@ def run_one(i, cfg_json, rng):
     with tempfile.TemporaryDirectory() as td:
         st = os.path.join(td, "m.safetensors")
         weights_to_safetensors(w, cfg, st)
+        # max_batch_tokens 32 forces CHUNKED prefill for most prompts
         eng = cake_amd.Engine(json.dumps(cfg_json), max_seq=256,
-                              max_batch_tokens=128)
+                              max_batch_tokens=32)
         eng.load_safetensors(st)
         try:
             first, lg = eng.prefill(prompt, want_logits=True)
